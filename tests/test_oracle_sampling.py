@@ -46,13 +46,17 @@ def test_sample_closest_rejects_bad_rate():
         sample_closest(np.arange(5, dtype=np.float32), 0.0)
 
 
-def test_sample_closest_2fps_10s_clip_is_20_frames():
-    """BASELINE workload shape: 10 s 30 fps clip sampled at 2 fps -> 20 frames."""
+def test_sample_closest_2fps_10s_clip_is_21_frames():
+    """BASELINE workload shape: 10 s 30 fps clip sampled at 2 fps.
+
+    The endpoint-epsilon rule (decoder_utils.py:364-371) includes the final
+    frame: grid 0.0..10.0 step 0.5 -> 21 samples, last lands on frame 299.
+    """
     ts = (np.arange(300) / 30.0).astype(np.float32)
     idx, counts, _ = sample_closest(ts, 2.0)
-    assert counts.sum() == 20
-    assert idx[0] == 0
-    assert np.all(np.diff(idx) == 15)
+    assert counts.sum() == 21
+    assert idx[0] == 0 and idx[-1] == 299
+    assert np.all(np.diff(idx[:-1]) == 15)
 
 
 def test_broadcast_selected_duplicates():
