@@ -1,0 +1,373 @@
+// Host MP4 (ISO BMFF) demuxer — the PTS + AnnexB packet source of the
+// rebuilt hot path.
+//
+// Replaces PyNvDemuxer (/root/reference/cosmos_curate/pipelines/video/utils/
+// nvcodec_utils.py:224) and implements the PTS contract of
+// get_video_timestamps (decoder_utils.py:230-278): per-sample presentation
+// timestamps from stts/ctts, shifted by the first non-empty elst edit
+// (libavformat mov semantics), converted to float32 seconds and sorted.
+//
+// Pure host C++ — runs without a GPU.  Parity: tests/test_demux_abi.py
+// checks this against oracle/mp4_demux.py on the committed fixtures.
+
+#include <algorithm>
+#include <cstring>
+#include <vector>
+
+#include "cc_common.hpp"
+
+namespace {
+
+struct Box {
+  char type[5];
+  size_t body_start, body_end;
+};
+
+// iterate boxes in data[start,end)
+struct BoxIter {
+  const uint8_t* data;
+  size_t pos, end;
+  bool next(Box* out) {
+    while (pos + 8 <= end) {
+      uint64_t size = (uint64_t)((data[pos] << 24) | (data[pos + 1] << 16) |
+                                 (data[pos + 2] << 8) | data[pos + 3]);
+      size_t hdr = 8;
+      if (size == 1) {
+        if (pos + 16 > end) return false;
+        size = 0;
+        for (int i = 0; i < 8; i++) size = (size << 8) | data[pos + 8 + i];
+        hdr = 16;
+      } else if (size == 0) {
+        size = end - pos;
+      }
+      if (size < hdr || pos + size > end) return false;
+      memcpy(out->type, data + pos + 4, 4);
+      out->type[4] = 0;
+      out->body_start = pos + hdr;
+      out->body_end = pos + size;
+      pos += size;
+      return true;
+    }
+    return false;
+  }
+};
+
+inline uint32_t rd32(const uint8_t* p) {
+  return ((uint32_t)p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
+}
+inline int32_t rd32s(const uint8_t* p) { return (int32_t)rd32(p); }
+inline uint16_t rd16(const uint8_t* p) { return (uint16_t)((p[0] << 8) | p[1]); }
+inline uint64_t rd64(const uint8_t* p) {
+  return ((uint64_t)rd32(p) << 32) | rd32(p + 4);
+}
+
+}  // namespace
+
+struct cc_demux {
+  std::vector<uint8_t> data;
+  // sample tables (first video track)
+  uint32_t timescale = 0;
+  int32_t codec = -1;  // 0 h264, 1 hevc
+  uint32_t width = 0, height = 0;
+  std::vector<uint32_t> sizes;
+  std::vector<uint64_t> offsets;
+  std::vector<int64_t> dts;
+  std::vector<int32_t> cts;
+  std::vector<uint32_t> sync;  // 1-based sample numbers; empty = all sync
+  int64_t elst_media_time = 0;
+  std::vector<uint8_t> avcc;  // avcC / hvcC record
+  int nal_len_size = 4;
+  std::vector<uint8_t> ps_prefix;  // SPS/PPS AnnexB prefix
+  std::vector<uint8_t> pkt_buf;    // scratch for cc_demux_packet
+};
+
+static bool parse_stbl(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
+  std::vector<std::pair<uint32_t, uint32_t>> stsc;  // first_chunk, samples_per_chunk
+  std::vector<uint64_t> chunk_offsets;
+  BoxIter it{d, b, e};
+  Box bx;
+  while (it.next(&bx)) {
+    const uint8_t* p = d + bx.body_start;
+    if (!strcmp(bx.type, "stsd")) {
+      uint32_t n = rd32(p + 4);
+      size_t pos = bx.body_start + 8;
+      for (uint32_t i = 0; i < n && pos + 8 <= bx.body_end; i++) {
+        uint32_t esize = rd32(d + pos);
+        char fmt[5] = {0};
+        memcpy(fmt, d + pos + 4, 4);
+        if (!strcmp(fmt, "avc1") || !strcmp(fmt, "avc3")) t->codec = 0;
+        if (!strcmp(fmt, "hvc1") || !strcmp(fmt, "hev1")) t->codec = 1;
+        if (t->codec >= 0) {
+          t->width = rd16(d + pos + 32);
+          t->height = rd16(d + pos + 34);
+          BoxIter it2{d, pos + 86, pos + esize};
+          Box b2;
+          while (it2.next(&b2)) {
+            if (!strcmp(b2.type, "avcC") || !strcmp(b2.type, "hvcC"))
+              t->avcc.assign(d + b2.body_start, d + b2.body_end);
+          }
+        }
+        pos += esize;
+      }
+    } else if (!strcmp(bx.type, "stts")) {
+      uint32_t n = rd32(p + 4);
+      int64_t tcur = 0;
+      size_t pos = bx.body_start + 8;
+      for (uint32_t i = 0; i < n; i++, pos += 8) {
+        uint32_t cnt = rd32(d + pos), delta = rd32(d + pos + 4);
+        for (uint32_t j = 0; j < cnt; j++) {
+          t->dts.push_back(tcur);
+          tcur += delta;
+        }
+      }
+    } else if (!strcmp(bx.type, "ctts")) {
+      uint32_t n = rd32(p + 4);
+      size_t pos = bx.body_start + 8;
+      for (uint32_t i = 0; i < n; i++, pos += 8) {
+        uint32_t cnt = rd32(d + pos);
+        int32_t off = rd32s(d + pos + 4);
+        for (uint32_t j = 0; j < cnt; j++) t->cts.push_back(off);
+      }
+    } else if (!strcmp(bx.type, "stss")) {
+      uint32_t n = rd32(p + 4);
+      for (uint32_t i = 0; i < n; i++) t->sync.push_back(rd32(p + 8 + 4 * i));
+    } else if (!strcmp(bx.type, "stsz")) {
+      uint32_t ss = rd32(p + 4), n = rd32(p + 8);
+      if (ss)
+        t->sizes.assign(n, ss);
+      else
+        for (uint32_t i = 0; i < n; i++) t->sizes.push_back(rd32(p + 12 + 4 * i));
+    } else if (!strcmp(bx.type, "stsc")) {
+      uint32_t n = rd32(p + 4);
+      for (uint32_t i = 0; i < n; i++)
+        stsc.push_back({rd32(p + 8 + 12 * i), rd32(p + 12 + 12 * i)});
+    } else if (!strcmp(bx.type, "stco") || !strcmp(bx.type, "co64")) {
+      uint32_t n = rd32(p + 4);
+      bool w = !strcmp(bx.type, "co64");
+      for (uint32_t i = 0; i < n; i++)
+        chunk_offsets.push_back(w ? rd64(p + 8 + 8 * i) : rd32(p + 8 + 4 * i));
+    }
+  }
+  if (t->cts.empty()) t->cts.assign(t->dts.size(), 0);
+  // expand chunk map to per-sample offsets
+  if (!chunk_offsets.empty() && !stsc.empty() && !t->sizes.empty()) {
+    std::vector<uint32_t> per_chunk;
+    for (size_t i = 0; i < stsc.size(); i++) {
+      uint32_t last = (i + 1 < stsc.size()) ? stsc[i + 1].first - 1
+                                            : (uint32_t)chunk_offsets.size();
+      for (uint32_t c = stsc[i].first; c <= last; c++)
+        per_chunk.push_back(stsc[i].second);
+    }
+    size_t si = 0;
+    for (size_t ci = 0; ci < chunk_offsets.size(); ci++) {
+      uint64_t pos = chunk_offsets[ci];
+      uint32_t cnt = ci < per_chunk.size() ? per_chunk[ci] : 0;
+      for (uint32_t j = 0; j < cnt && si < t->sizes.size(); j++, si++) {
+        t->offsets.push_back(pos);
+        pos += t->sizes[si];
+      }
+    }
+  }
+  return true;
+}
+
+static bool parse_trak(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
+  bool is_video = false;
+  BoxIter it{d, b, e};
+  Box bx;
+  while (it.next(&bx)) {
+    const uint8_t* p = d + bx.body_start;
+    if (!strcmp(bx.type, "edts")) {
+      BoxIter it2{d, bx.body_start, bx.body_end};
+      Box b2;
+      while (it2.next(&b2)) {
+        if (!strcmp(b2.type, "elst")) {
+          const uint8_t* q = d + b2.body_start;
+          uint8_t ver = q[0];
+          uint32_t n = rd32(q + 4);
+          size_t pos = b2.body_start + 8;
+          for (uint32_t i = 0; i < n; i++) {
+            int64_t media_time;
+            if (ver == 1) {
+              media_time = (int64_t)rd64(d + pos + 8);
+              pos += 20;
+            } else {
+              media_time = rd32s(d + pos + 4);
+              pos += 12;
+            }
+            if (media_time != -1) {
+              t->elst_media_time = media_time;
+              break;
+            }
+          }
+        }
+      }
+    } else if (!strcmp(bx.type, "mdia")) {
+      BoxIter it2{d, bx.body_start, bx.body_end};
+      Box b2;
+      while (it2.next(&b2)) {
+        const uint8_t* q = d + b2.body_start;
+        if (!strcmp(b2.type, "mdhd")) {
+          uint8_t ver = q[0];
+          t->timescale = rd32(q + (ver == 1 ? 20 : 12));
+        } else if (!strcmp(b2.type, "hdlr")) {
+          if (!memcmp(q + 8, "vide", 4)) is_video = true;
+        } else if (!strcmp(b2.type, "minf")) {
+          BoxIter it3{d, b2.body_start, b2.body_end};
+          Box b3;
+          while (it3.next(&b3))
+            if (!strcmp(b3.type, "stbl")) parse_stbl(d, b3.body_start, b3.body_end, t);
+        }
+      }
+    }
+  }
+  return is_video && t->timescale > 0;
+}
+
+static void build_ps_prefix(cc_demux* t) {
+  // AnnexB parameter-set prefix from avcC (h264 only for now)
+  if (t->codec != 0 || t->avcc.size() < 7) return;
+  const uint8_t* a = t->avcc.data();
+  t->nal_len_size = (a[4] & 0x03) + 1;
+  size_t pos = 5;
+  auto append_sets = [&](int count) {
+    static const uint8_t sc[4] = {0, 0, 0, 1};
+    for (int i = 0; i < count && pos + 2 <= t->avcc.size(); i++) {
+      uint16_t ln = rd16(a + pos);
+      pos += 2;
+      if (pos + ln > t->avcc.size()) return;
+      t->ps_prefix.insert(t->ps_prefix.end(), sc, sc + 4);
+      t->ps_prefix.insert(t->ps_prefix.end(), a + pos, a + pos + ln);
+      pos += ln;
+    }
+  };
+  int num_sps = a[5] & 0x1F;
+  pos = 6;  // skip the SPS-count byte
+  append_sets(num_sps);
+  if (pos < t->avcc.size()) {
+    int num_pps = a[pos++];
+    append_sets(num_pps);
+  }
+}
+
+extern "C" {
+
+const char* cc_last_error(void) { return cc::g_last_error.c_str(); }
+
+int cc_demux_open(const uint8_t* data, size_t size, cc_demux** out) {
+  if (!data || !size || !out) return cc::set_error(CC_ERR_INVALID, "null arg");
+  auto* t = new cc_demux();
+  t->data.assign(data, data + size);
+  const uint8_t* d = t->data.data();
+  BoxIter it{d, 0, size};
+  Box bx;
+  bool found = false;
+  while (it.next(&bx) && !found) {
+    if (!strcmp(bx.type, "moov")) {
+      BoxIter it2{d, bx.body_start, bx.body_end};
+      Box b2;
+      while (it2.next(&b2)) {
+        if (!strcmp(b2.type, "trak")) {
+          cc_demux probe;
+          if (parse_trak(d, b2.body_start, b2.body_end, &probe) &&
+              probe.codec >= 0) {
+            // keep tables, move into t
+            t->timescale = probe.timescale;
+            t->codec = probe.codec;
+            t->width = probe.width;
+            t->height = probe.height;
+            t->sizes = std::move(probe.sizes);
+            t->offsets = std::move(probe.offsets);
+            t->dts = std::move(probe.dts);
+            t->cts = std::move(probe.cts);
+            t->sync = std::move(probe.sync);
+            t->elst_media_time = probe.elst_media_time;
+            t->avcc = std::move(probe.avcc);
+            found = true;
+            break;
+          }
+        }
+      }
+    }
+  }
+  if (!found || t->dts.empty()) {
+    delete t;
+    return cc::set_error(CC_ERR_PARSE, "no decodable video track");
+  }
+  build_ps_prefix(t);
+  *out = t;
+  return CC_OK;
+}
+
+int cc_demux_probe(const cc_demux* d, cc_video_info* info) {
+  if (!d || !info) return cc::set_error(CC_ERR_INVALID, "null arg");
+  info->width = d->width;
+  info->height = d->height;
+  info->timescale = d->timescale;
+  info->num_samples = (uint32_t)d->dts.size();
+  info->num_sync_samples = (uint32_t)d->sync.size();
+  info->codec = d->codec;
+  int64_t span = d->dts.empty() ? 0 : d->dts.back() - d->dts.front();
+  // media duration: last dts + last delta ~= samples * avg delta
+  double dur = d->dts.size() > 1
+                   ? (double)span / (d->dts.size() - 1) * d->dts.size() / d->timescale
+                   : 0.0;
+  info->duration_s = dur;
+  info->avg_fps = dur > 0 ? d->dts.size() / dur : 0.0;
+  return CC_OK;
+}
+
+int cc_demux_timestamps(const cc_demux* d, float* out, size_t cap, size_t* n) {
+  if (!d || !n) return cc::set_error(CC_ERR_INVALID, "null arg");
+  *n = d->dts.size();
+  if (!out) return CC_OK;  // size query
+  if (cap < d->dts.size()) return cc::set_error(CC_ERR_INVALID, "cap too small");
+  std::vector<float> ts(d->dts.size());
+  for (size_t i = 0; i < d->dts.size(); i++) {
+    // float32 of (pts - elst_shift) / timescale, exactly as
+    // decoder_utils.py:275 computes float(pts) * time_base in f64 then
+    // narrows to f32 on array construction.
+    double sec = (double)(d->dts[i] + d->cts[i] - d->elst_media_time) /
+                 (double)d->timescale;
+    ts[i] = (float)sec;
+  }
+  std::sort(ts.begin(), ts.end());
+  memcpy(out, ts.data(), ts.size() * sizeof(float));
+  return CC_OK;
+}
+
+int cc_demux_packet(cc_demux* d, size_t index, const uint8_t** pkt, size_t* size,
+                    int64_t* pts, int32_t* keyframe) {
+  if (!d || !pkt || !size) return cc::set_error(CC_ERR_INVALID, "null arg");
+  if (index >= d->sizes.size() || index >= d->offsets.size())
+    return cc::set_error(CC_ERR_INVALID, "sample index out of range");
+  const uint8_t* sample = d->data.data() + d->offsets[index];
+  size_t ssize = d->sizes[index];
+  if (d->offsets[index] + ssize > d->data.size())
+    return cc::set_error(CC_ERR_PARSE, "sample range outside file");
+  bool is_sync = d->sync.empty() ||
+                 std::binary_search(d->sync.begin(), d->sync.end(), (uint32_t)(index + 1));
+  d->pkt_buf.clear();
+  if (is_sync)
+    d->pkt_buf.insert(d->pkt_buf.end(), d->ps_prefix.begin(), d->ps_prefix.end());
+  static const uint8_t sc[4] = {0, 0, 0, 1};
+  size_t p = 0;
+  while (p + d->nal_len_size <= ssize) {
+    uint64_t ln = 0;
+    for (int i = 0; i < d->nal_len_size; i++) ln = (ln << 8) | sample[p + i];
+    p += d->nal_len_size;
+    size_t take = std::min<uint64_t>(ln, ssize - p);
+    d->pkt_buf.insert(d->pkt_buf.end(), sc, sc + 4);
+    d->pkt_buf.insert(d->pkt_buf.end(), sample + p, sample + p + take);
+    p += take;
+  }
+  *pkt = d->pkt_buf.data();
+  *size = d->pkt_buf.size();
+  if (pts) *pts = d->dts[index] + d->cts[index] - d->elst_media_time;
+  if (keyframe) *keyframe = is_sync ? 1 : 0;
+  return CC_OK;
+}
+
+void cc_demux_close(cc_demux* d) { delete d; }
+
+}  // extern "C"

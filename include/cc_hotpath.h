@@ -1,0 +1,154 @@
+/* cc_hotpath.h — C ABI of the MI355X-native cosmos-curate hot path.
+ *
+ * One shared library (libcchot.so, built from cosmos_curate_amd/csrc/ by
+ * hipcc for gfx950) exporting the functionality the reference obtains from
+ * external native libraries.  Each entry point cites
+ * the reference interface it replaces (file:line relative to
+ * /root/reference/cosmos_curate/).  Plain pointers + sizes only; device
+ * memory is referenced by raw HIP device pointers allocated either by the
+ * caller's framework (torch) or by cc_malloc below; `stream` arguments are
+ * hipStream_t handles passed as uint64 (0 = default stream).
+ *
+ * Host-side entry points (cc_demux_*) never touch the GPU and work in a
+ * GPU-less container; every cc_* device function requires a visible GPU and
+ * fails with CC_ERR_HIP otherwise (no CPU fallback anywhere).
+ *
+ * Errors: every function returns 0 on success or a negative CC_ERR_* code;
+ * cc_last_error() returns a thread-local message for the last failure.
+ * Thread-safety: handles are single-threaded; distinct handles independent.
+ */
+
+#ifndef CC_HOTPATH_H
+#define CC_HOTPATH_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+enum {
+  CC_OK = 0,
+  CC_ERR_INVALID = -1,     /* bad argument */
+  CC_ERR_PARSE = -2,       /* malformed container */
+  CC_ERR_NOMEM = -3,
+  CC_ERR_HIP = -4,         /* HIP runtime failure (incl. no GPU) */
+  CC_ERR_NO_ROCDECODE = -5,/* VCN decode library not present at runtime */
+  CC_ERR_UNSUPPORTED = -6,
+};
+
+const char* cc_last_error(void);
+
+/* ---- version / capability probes ---------------------------------- */
+/* Returns CC_OK when a HIP device is visible and usable. */
+int cc_hip_available(void);
+/* Returns CC_OK when librocdecode can be dlopened (VCN decode usable). */
+int cc_rocdecode_available(void);
+
+/* ---- host MP4 demux ------------------------------------------------
+ * Replaces PyNvDemuxer (nvcodec_utils.py:224, fps at :124) and is the PTS
+ * source behind get_video_timestamps (pipelines/video/utils/
+ * decoder_utils.py:230-278).  Pure host C++; no codec work. */
+typedef struct cc_demux cc_demux_t;
+
+typedef struct cc_video_info {
+  uint32_t width, height;
+  uint32_t timescale;
+  uint32_t num_samples;
+  uint32_t num_sync_samples;
+  int32_t  codec;           /* 0 = h264(avc1/avc3), 1 = hevc(hvc1/hev1) */
+  double   duration_s;      /* track media duration in seconds */
+  double   avg_fps;         /* num_samples / duration */
+} cc_video_info;
+
+int  cc_demux_open(const uint8_t* data, size_t size, cc_demux_t** out);
+int  cc_demux_probe(const cc_demux_t* d, cc_video_info* info);
+/* Sorted presentation timestamps in seconds, float32 — the exact
+ * get_video_timestamps contract (decoder_utils.py:275-278).  `cap` is the
+ * capacity of `out`; *n gets the sample count. */
+int  cc_demux_timestamps(const cc_demux_t* d, float* out, size_t cap, size_t* n);
+/* AnnexB packet for sample `index` (decode order): start-code NALs, with
+ * SPS/PPS prefixed on sync samples.  Buffer owned by the demuxer, valid
+ * until the next cc_demux_packet call or close. */
+int  cc_demux_packet(cc_demux_t* d, size_t index, const uint8_t** pkt,
+                     size_t* size, int64_t* pts, int32_t* keyframe);
+void cc_demux_close(cc_demux_t* d);
+
+/* ---- device memory (library-owned buffers; callers may also pass
+ * torch-allocated device pointers to the kernels below) ---------------- */
+int cc_malloc(void** dptr, size_t bytes);
+int cc_free(void* dptr);
+int cc_memcpy_h2d(void* dst, const void* src, size_t bytes, uint64_t stream);
+int cc_memcpy_d2h(void* dst, const void* src, size_t bytes, uint64_t stream);
+int cc_stream_sync(uint64_t stream);
+
+/* ---- VCN hardware decode (rocDecode, runtime-probed) ----------------
+ * Replaces NVDEC via PyNvVideoCodec CreateDecoder/Decode
+ * (nvcodec_utils.py:199-313).  Fails with CC_ERR_NO_ROCDECODE when
+ * librocdecode.so is absent (this image ships none; the ABI is the seam). */
+typedef struct cc_decode cc_decode_t;
+int  cc_decode_session_create(int device, int32_t codec, cc_decode_t** out);
+int  cc_decode_submit(cc_decode_t* s, const uint8_t* pkt, size_t size, int64_t pts);
+/* Mapped NV12 surfaces land as {y_ptr, uv_ptr, pitch} triples. */
+typedef struct cc_nv12_frame {
+  void* y; void* uv; size_t pitch; int64_t pts; uint32_t width, height;
+} cc_nv12_frame;
+int  cc_decode_map_frames(cc_decode_t* s, cc_nv12_frame* out, size_t cap, size_t* n);
+void cc_decode_destroy(cc_decode_t* s);
+
+/* ---- fused pixel kernels (hand-written HIP, gfx950) -----------------
+ * Replace cvcuda.cvtcolor_into(YUV2RGB_NV12) (nvcodec_utils.py:178),
+ * cvcuda.resize_into(LINEAR) (:189-194), cvcuda.reformat_into (:267),
+ * cv2 INTER_CUBIC (decoder_utils.py:666-670) and the CLIP/torchvision
+ * normalize chain (models/clip.py:48-62).  Batched NHWC u8 layouts. */
+
+/* NV12 (Y plane + interleaved UV half-res plane, same pitch) ->
+ * RGB888 NHWC at (out_h,out_w) via bilinear taps in converted-RGB space
+ * (each tap converted BT.601 limited-range and rounded to u8 first —
+ * bit-identical to convert-then-resize, fused to skip the intermediate). */
+int cc_nv12_to_rgb_resize(const void* y, const void* uv, int n,
+                          int src_h, int src_w, size_t pitch,
+                          void* out_rgb, int out_h, int out_w,
+                          uint64_t stream);
+/* Plain NV12 -> RGB888 full resolution (the cvtcolor_into twin). */
+int cc_nv12_to_rgb(const void* y, const void* uv, int n,
+                   int h, int w, size_t pitch, void* out_rgb, uint64_t stream);
+/* u8 NHWC bilinear resize (cvcuda LINEAR semantics). */
+int cc_resize_bilinear_u8(const void* in, int n, int src_h, int src_w,
+                          void* out, int dst_h, int dst_w, uint64_t stream);
+/* u8 NHWC bicubic resize, A=-0.75 (cv2 INTER_CUBIC semantics). */
+int cc_resize_bicubic_u8(const void* in, int n, int src_h, int src_w,
+                         void* out, int dst_h, int dst_w, uint64_t stream);
+/* (N,H,W,3) u8 -> (N,3,H,W) normalized ((x/255)-mean)/std.
+ * out_dtype: 0 = f32, 1 = bf16. */
+int cc_clip_preprocess(const void* in, int n, int h, int w,
+                       const float mean[3], const float stdev[3],
+                       void* out, int out_dtype, uint64_t stream);
+/* Gather + duplicate-count broadcast of selected frames on device
+ * (the decode loop's count broadcast, decoder_utils.py:447-453). */
+int cc_gather_frames_u8(const void* frames, int n_in, size_t frame_bytes,
+                        const int32_t* idx, const int32_t* counts, int n_idx,
+                        int total_out, void* out, uint64_t stream);
+
+/* ---- ViT MFMA GEMMs -------------------------------------------------
+ * Replace the cuBLAS GEMMs under CLIPModel.get_image_features
+ * (models/clip.py:71): patch-embed-as-GEMM, QKV/out projections, MLP.
+ * C[M,N] = A[M,K] (bf16, row-major) x B[N,K]^T (bf16, row-major "weight
+ * layout") + bias[N] (f32, optional NULL).  c_dtype: 0 = f32, 1 = bf16. */
+int cc_gemm_bf16(const void* A, const void* B, void* C,
+                 int64_t M, int64_t N, int64_t K,
+                 const float* bias, int c_dtype, uint64_t stream);
+
+/* ---- kernel timing (bench.py roofline evidence) ---------------------
+ * When enabled, every cc_* kernel launch is bracketed with hipEvents on
+ * its launch stream; totals are accumulated per kernel name. */
+int cc_timing_enable(int enable);
+int cc_timing_reset(void);
+/* Fetch totals for `kernel` (e.g. "gemm_bf16"): total device ms + count. */
+int cc_timing_report(const char* kernel, double* total_ms, int64_t* count);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* CC_HOTPATH_H */
