@@ -1,0 +1,99 @@
+"""CLIPImageEmbeddings ModelInterface — the rebuild's flagship embedder.
+
+Mirror of /root/reference/cosmos_curate/models/clip.py:36-118: same class
+name, same ModelInterface surface (conda_env_name / model_id_names /
+setup / __call__ over u8 NHWC frames or NCHW tensors), same output contract
+(L2-normalized image embeddings, clip.py:71-74).  Inside: the CLIP
+preprocess chain runs as one fused HIP kernel (cc_clip_preprocess) and the
+ViT forward on the MFMA GEMM path (clip_vit.ClipVisionTowerAMD), bf16.
+
+At the benchmark config frames arrive already resized to 224x224
+(clip_frame_extraction target_res, SURVEY.md §2 row "Clip frame
+extraction"), so torchvision's Resize/CenterCrop (clip.py:48-56) are
+identity; inputs of other sizes are resized on device with the bicubic
+kernel first (torchvision BICUBIC semantics differ from cv2 INTER_CUBIC
+only in antialias, which is OFF for upscaling and irrelevant at 224->224).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import numpy.typing as npt
+import torch
+
+from cosmos_curate_amd import hotpath
+from cosmos_curate_amd.core.interfaces.model_interface import ModelInterface
+from cosmos_curate_amd.models.clip_vit import ClipVisionTowerAMD
+from cosmos_curate_amd.models.clip_weights import make_clip_vit_b32_weights
+
+_CLIP_MODEL_ID = "openai/clip-vit-base-patch32"
+
+CLIP_MEAN = (0.48145466, 0.4578275, 0.40821073)
+CLIP_STD = (0.26862954, 0.26130258, 0.27577711)
+
+
+class _CLIPImageEmbeddings(torch.nn.Module):
+    """Device-side implementation (reference clip.py:36-74 counterpart)."""
+
+    def __init__(self) -> None:
+        super().__init__()
+        hotpath.require_gpu()  # fail loudly before any lazy surprises
+        self.device = torch.device("cuda")
+        self.tower = ClipVisionTowerAMD(make_clip_vit_b32_weights()).to(self.device)
+        self._mean = (np.array(CLIP_MEAN, dtype=np.float32) * 1.0).ctypes
+        self._std = (np.array(CLIP_STD, dtype=np.float32) * 1.0).ctypes
+        self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
+        self._std_arr = np.array(CLIP_STD, dtype=np.float32)
+
+    def preprocess_u8(self, frames_dev_u8: torch.Tensor) -> torch.Tensor:
+        """(N,224,224,3) u8 on device -> (N,3,224,224) bf16 normalized."""
+        import ctypes
+
+        lib = hotpath.require_gpu()
+        n, h, w, _ = frames_dev_u8.shape
+        out = torch.empty((n, 3, h, w), dtype=torch.bfloat16, device=frames_dev_u8.device)
+        stream = torch.cuda.current_stream(frames_dev_u8.device).cuda_stream
+        mean = (ctypes.c_float * 3)(*self._mean_arr)
+        std = (ctypes.c_float * 3)(*self._std_arr)
+        hotpath.check(
+            lib.cc_clip_preprocess(
+                frames_dev_u8.contiguous().data_ptr(), n, h, w, mean, std,
+                out.data_ptr(), 1, stream,
+            )
+        )
+        return out
+
+    @torch.no_grad()
+    def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
+        if isinstance(images, np.ndarray):
+            # (N,H,W,C) u8 host array (reference clip.py:66-68 entry form)
+            dev = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
+            pixels = self.preprocess_u8(dev)
+        elif images.dtype == torch.uint8 and images.ndim == 4 and images.shape[-1] == 3:
+            pixels = self.preprocess_u8(images.to(self.device))
+        else:
+            pixels = images.to(self.device, dtype=torch.bfloat16)
+        return self.tower(pixels)
+
+
+class CLIPImageEmbeddings(ModelInterface):
+    """ModelInterface wrapper (reference clip.py:77-118)."""
+
+    def __init__(self) -> None:
+        super().__init__()
+        self._model: _CLIPImageEmbeddings | None = None
+
+    @property
+    def conda_env_name(self) -> str:
+        return "unified"  # constant on the rebuild (single ROCm env)
+
+    @property
+    def model_id_names(self) -> list[str]:
+        return [_CLIP_MODEL_ID]
+
+    def setup(self) -> None:
+        self._model = _CLIPImageEmbeddings()
+
+    def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
+        assert self._model is not None, "setup() not called"
+        return self._model(images)

@@ -1,0 +1,156 @@
+"""CLIP ViT-B/32 vision tower, MI355X-native forward.
+
+Product counterpart of ``CLIPModel.get_image_features``
+(/root/reference/cosmos_curate/models/clip.py:64-74): same arithmetic as
+transformers' CLIPVisionModelWithProjection (conv patch embed -> cls+pos ->
+pre-LN -> 12 x [LN1, MHA, LN2, quick-gelu MLP] -> post-LN on cls ->
+visual projection), organized the MI355X way:
+
+- every big contraction (patch-embed-as-GEMM, fused QKV, out-proj,
+  fc1/fc2, visual projection) runs on the hand-written MFMA bf16 kernel
+  (cc_gemm_bf16, csrc/cc_gemm.hip) through the C ABI;
+- attention softmax/bmm and layernorms stay on torch-rocm
+  (north_star: "the rest stays torch-rocm"); layernorms compute in f32;
+- activations are bf16 end-to-end; the final embedding is cast to f32 and
+  L2-normalized (clip.py:74).
+
+No CPU fallback: _linear() requires the HIP extension + a GPU (the tests
+may monkeypatch _linear to torch on CPU to pin arithmetic structure; the
+product never does).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from cosmos_curate_amd import hotpath
+from cosmos_curate_amd.models import clip_weights as cw
+
+
+def _cc_linear(x: torch.Tensor, w_bf16: torch.Tensor, bias_f32: torch.Tensor | None) -> torch.Tensor:
+    """C[M,N] = x[M,K] @ w[N,K]^T + bias via cc_gemm_bf16 (bf16 out)."""
+    lib = hotpath.require_gpu()
+    assert x.dtype == torch.bfloat16 and w_bf16.dtype == torch.bfloat16
+    x = x.contiguous()
+    M, K = x.shape
+    N = w_bf16.shape[0]
+    out = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
+    stream = torch.cuda.current_stream(x.device).cuda_stream
+    hotpath.check(
+        lib.cc_gemm_bf16(
+            x.data_ptr(), w_bf16.data_ptr(), out.data_ptr(), M, N, K,
+            bias_f32.data_ptr() if bias_f32 is not None else None,
+            1, stream,
+        )
+    )
+    return out
+
+
+class ClipVisionTowerAMD(torch.nn.Module):
+    """ViT-B/32 vision tower + projection on the MFMA GEMM path."""
+
+    def __init__(self, state_dict: dict[str, torch.Tensor] | None = None) -> None:
+        super().__init__()
+        sd = state_dict if state_dict is not None else cw.make_clip_vit_b32_weights()
+        p = "vision_model."
+        reg = self.register_buffer
+        to_bf = lambda t: t.to(torch.bfloat16).contiguous()  # noqa: E731
+
+        reg("cls_emb", sd[p + "embeddings.class_embedding"].clone())
+        # patch conv as GEMM weight: [768, 3*32*32] with (c,ky,kx) flattening
+        reg("w_patch", to_bf(sd[p + "embeddings.patch_embedding.weight"].reshape(cw.HIDDEN, -1)))
+        reg("pos_emb", sd[p + "embeddings.position_embedding.weight"].clone())
+        reg("pre_ln_w", sd[p + "pre_layrnorm.weight"].clone())
+        reg("pre_ln_b", sd[p + "pre_layrnorm.bias"].clone())
+        self.layers = len(
+            {k.split(".")[3] for k in sd if k.startswith(p + "encoder.layers.")}
+        )
+        for i in range(self.layers):
+            q = f"{p}encoder.layers.{i}."
+            # fused QKV: [3*768, 768] rows ordered (q, k, v)
+            wq, wk, wv = (sd[q + f"self_attn.{x}.weight"] for x in ("q_proj", "k_proj", "v_proj"))
+            bq, bk, bv = (sd[q + f"self_attn.{x}.bias"] for x in ("q_proj", "k_proj", "v_proj"))
+            reg(f"w_qkv_{i}", to_bf(torch.cat([wq, wk, wv], dim=0)))
+            reg(f"b_qkv_{i}", torch.cat([bq, bk, bv]).float().contiguous())
+            reg(f"w_out_{i}", to_bf(sd[q + "self_attn.out_proj.weight"]))
+            reg(f"b_out_{i}", sd[q + "self_attn.out_proj.bias"].float().contiguous())
+            reg(f"ln1_w_{i}", sd[q + "layer_norm1.weight"].clone())
+            reg(f"ln1_b_{i}", sd[q + "layer_norm1.bias"].clone())
+            reg(f"ln2_w_{i}", sd[q + "layer_norm2.weight"].clone())
+            reg(f"ln2_b_{i}", sd[q + "layer_norm2.bias"].clone())
+            reg(f"w_fc1_{i}", to_bf(sd[q + "mlp.fc1.weight"]))
+            reg(f"b_fc1_{i}", sd[q + "mlp.fc1.bias"].float().contiguous())
+            reg(f"w_fc2_{i}", to_bf(sd[q + "mlp.fc2.weight"]))
+            reg(f"b_fc2_{i}", sd[q + "mlp.fc2.bias"].float().contiguous())
+        reg("post_ln_w", sd[p + "post_layernorm.weight"].clone())
+        reg("post_ln_b", sd[p + "post_layernorm.bias"].clone())
+        reg("w_proj", to_bf(sd["visual_projection.weight"]))
+        self.heads = cw.HEADS
+        self.scale = 1.0 / math.sqrt(cw.HIDDEN // cw.HEADS)
+
+    # the one contraction primitive; tests may monkeypatch this
+    def _linear(self, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None) -> torch.Tensor:
+        return _cc_linear(x, w, b)
+
+    def _ln(self, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        return torch.nn.functional.layer_norm(
+            x.float(), (x.shape[-1],), w, b, eps=1e-5
+        ).to(x.dtype)
+
+    @torch.no_grad()
+    def forward(self, pixel_values: torch.Tensor) -> torch.Tensor:
+        """(N,3,224,224) bf16 -> (N,512) f32 L2-normalized embeddings."""
+        x = pixel_values.to(torch.bfloat16)
+        n = x.shape[0]
+        g = cw.IMAGE // cw.PATCH  # 7
+        # patch extraction: (n,c,ph,ky,pw,kx) -> (n,ph,pw,c,ky,kx)
+        patches = (
+            x.reshape(n, 3, g, cw.PATCH, g, cw.PATCH)
+            .permute(0, 2, 4, 1, 3, 5)
+            .reshape(n * g * g, 3 * cw.PATCH * cw.PATCH)
+        )
+        tok = self._linear(patches, self.w_patch, None).reshape(n, g * g, cw.HIDDEN)
+        cls = self.cls_emb.to(tok.dtype).expand(n, 1, cw.HIDDEN)
+        h = torch.cat([cls, tok], dim=1)  # (n, 50, 768)
+        h = (h.float() + self.pos_emb.unsqueeze(0)).to(torch.bfloat16)
+        h = self._ln(h, self.pre_ln_w, self.pre_ln_b)
+
+        seq = h.shape[1]
+        hd = cw.HIDDEN // self.heads
+        for i in range(self.layers):
+            res = h
+            y = self._ln(h, getattr(self, f"ln1_w_{i}"), getattr(self, f"ln1_b_{i}"))
+            qkv = self._linear(
+                y.reshape(n * seq, cw.HIDDEN),
+                getattr(self, f"w_qkv_{i}"),
+                getattr(self, f"b_qkv_{i}"),
+            ).reshape(n, seq, 3, self.heads, hd)
+            q = qkv[:, :, 0].permute(0, 2, 1, 3)  # (n, heads, seq, hd)
+            k = qkv[:, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, 2].permute(0, 2, 1, 3)
+            attn = torch.nn.functional.scaled_dot_product_attention(
+                q, k, v, scale=self.scale
+            )
+            attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cw.HIDDEN)
+            h = res + self._linear(
+                attn, getattr(self, f"w_out_{i}"), getattr(self, f"b_out_{i}")
+            ).reshape(n, seq, cw.HIDDEN)
+
+            res = h
+            y = self._ln(h, getattr(self, f"ln2_w_{i}"), getattr(self, f"ln2_b_{i}"))
+            y = self._linear(
+                y.reshape(n * seq, cw.HIDDEN),
+                getattr(self, f"w_fc1_{i}"),
+                getattr(self, f"b_fc1_{i}"),
+            )
+            y = y * torch.sigmoid(1.702 * y)  # quick_gelu (transformers CLIP)
+            y = self._linear(
+                y, getattr(self, f"w_fc2_{i}"), getattr(self, f"b_fc2_{i}")
+            ).reshape(n, seq, cw.HIDDEN)
+            h = res + y
+
+        pooled = self._ln(h[:, 0], self.post_ln_w, self.post_ln_b)
+        emb = self._linear(pooled.to(torch.bfloat16), self.w_proj, None).float()
+        return emb / torch.linalg.vector_norm(emb, dim=-1, keepdim=True)

@@ -1,0 +1,141 @@
+"""Stage + runner integration on CPU (SequentialRunner, no GPU).
+
+Mirrors the reference's test approach: real stages driven by
+a SequentialRunner mini-pipeline; golden expectations from
+test_fixed_stride_extraction.py (30 s video -> 3 x 10 s clips, etc.).
+"""
+
+import pathlib
+import uuid
+
+import numpy as np
+import pytest
+
+from cosmos_curate_amd.core.interfaces import (
+    CuratorStage,
+    CuratorStageSpec,
+    PipelineExecutionError,
+    SequentialRunner,
+    WorkerPoolRunner,
+    run_pipeline,
+)
+from cosmos_curate_amd.pipelines.video.clipping.clip_extraction_stages import (
+    FixedStrideExtractorStage,
+)
+from cosmos_curate_amd.pipelines.video.utils.data_model import (
+    SplitPipeTask,
+    Video,
+    VideoMetadata,
+)
+
+
+def make_task(duration_s: float = 30.0, fps: float = 24.0) -> SplitPipeTask:
+    n = int(duration_s * fps)
+    v = Video(
+        input_video=pathlib.Path("/data/test_video.mp4"),
+        metadata=VideoMetadata(
+            size=1000, height=480, width=854, framerate=fps, num_frames=n,
+            duration=duration_s, video_codec="h264",
+        ),
+        timestamps=(np.arange(n) / fps).astype(np.float32),
+    )
+    return SplitPipeTask(videos=[v])
+
+
+def test_fixed_stride_default_parameters():
+    """test_fixed_stride_extraction.py:100-141 golden: 3 clips 0-10/10-20/20-30."""
+    stage = FixedStrideExtractorStage(log_stats=True)
+    assert stage.clip_len_s == 10 and stage.clip_stride_s == 10
+    assert stage.min_clip_length_s == 10 and stage._limit_clips == 0
+    assert stage.resources.cpus == 1.0 and stage.resources.gpus == 0
+
+    out = run_pipeline([make_task()], [stage], runner=SequentialRunner())
+    assert len(out) == 1
+    video = out[0].video
+    assert len(video.clips) == 3
+    expected = [(0.0, 10.0), (10.0, 20.0), (20.0, 30.0)]
+    for clip, (s, e) in zip(video.clips, expected):
+        assert clip.span[0] == pytest.approx(s, abs=0.01)
+        assert clip.span[1] == pytest.approx(e, abs=0.01)
+    assert "FixedStrideExtractorStage" in out[0].stage_perf
+
+
+def test_fixed_stride_5s():
+    """test_fixed_stride_extraction.py:146-186 golden: 6 clips of 5 s."""
+    stage = FixedStrideExtractorStage(
+        clip_len_s=5.0, clip_stride_s=5.0, min_clip_length_s=5.0
+    )
+    out = run_pipeline([make_task()], [stage], runner=SequentialRunner())
+    assert len(out[0].video.clips) == 6
+
+
+def test_fixed_stride_uuid_determinism():
+    """UUID formula: uuid5(NAMESPACE_URL, f"{session}_{s}_{e}") with
+    session = input path (test_fixed_stride_extraction.py:406 pattern)."""
+    out1 = run_pipeline([make_task()], [FixedStrideExtractorStage()], runner=SequentialRunner())
+    out2 = run_pipeline([make_task()], [FixedStrideExtractorStage()], runner=SequentialRunner())
+    u1 = [c.uuid for c in out1[0].video.clips]
+    u2 = [c.uuid for c in out2[0].video.clips]
+    assert u1 == u2
+    assert u1[0] == uuid.uuid5(uuid.NAMESPACE_URL, "/data/test_video.mp4_0.0_10.0")
+
+
+def test_invalid_duration_records_error_not_raise():
+    t = make_task()
+    t.video.metadata.num_frames = 0
+    out = run_pipeline([t], [FixedStrideExtractorStage()], runner=SequentialRunner())
+    assert out[0].video.errors  # recorded, not raised (§8b error convention)
+    assert not out[0].video.clips
+
+
+def test_run_pipeline_wraps_failures():
+    class Boom(CuratorStage):
+        def process_data(self, tasks):
+            raise RuntimeError("boom")
+
+    with pytest.raises(PipelineExecutionError):
+        run_pipeline([make_task()], [Boom()], runner=SequentialRunner())
+
+
+def test_spec_defaults_gpu_lifetime():
+    class G(CuratorStage):
+        @property
+        def resources(self):
+            from cosmos_curate_amd.core.interfaces import CuratorStageResource
+
+            return CuratorStageResource(gpus=1)
+
+    from cosmos_curate_amd.core.interfaces.pipeline_interface import (
+        _build_pipeline_stage_specs,
+    )
+
+    specs = _build_pipeline_stage_specs([G()])
+    assert specs[0].worker_max_lifetime_m == 120  # pipeline_interface.py:187-219
+    assert specs[0].worker_restart_interval_m == 5
+
+
+def test_worker_pool_runner_shards():
+    tasks = [make_task() for _ in range(7)]
+    r0 = WorkerPoolRunner(rank=0, world_size=2)
+    r1 = WorkerPoolRunner(rank=1, world_size=2)
+    s0 = r0.shard(tasks)
+    s1 = r1.shard(tasks)
+    assert len(s0) == 4 and len(s1) == 3
+    assert {id(t) for t in s0} | {id(t) for t in s1} == {id(t) for t in tasks}
+    out = r0.run(tasks, [CuratorStageSpec(FixedStrideExtractorStage())])
+    assert len(out) == 4
+
+
+def test_retry_attempts():
+    calls = {"n": 0}
+
+    class Flaky(CuratorStage):
+        def process_data(self, tasks):
+            calls["n"] += 1
+            if calls["n"] < 3:
+                raise RuntimeError("transient")
+            return tasks
+
+    spec = CuratorStageSpec(Flaky(), num_run_attempts_python=3)
+    out = run_pipeline([make_task()], [spec], runner=SequentialRunner())
+    assert calls["n"] == 3 and len(out) == 1

@@ -1,0 +1,56 @@
+"""Product ViT arithmetic vs the transformers oracle (CPU, no GPU).
+
+Pins the *structure* of ClipVisionTowerAMD (patch-GEMM flattening, fused
+QKV ordering, quick-gelu, LN placement, projection, L2 norm) against the
+reference's own arithmetic (transformers CLIPVisionModelWithProjection,
+models/clip.py:64-74) by monkeypatching the one contraction primitive
+(_linear) to torch on CPU.  The MFMA kernel itself is pinned on the GPU in
+tests/test_gpu_vit.py; this test makes a GPU failure attributable to the
+kernel alone.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from cosmos_curate_amd.models.clip_vit import ClipVisionTowerAMD
+from cosmos_curate_amd.models.clip_weights import make_clip_vit_b32_weights
+from oracle import vit as oracle_vit
+from oracle.color import clip_preprocess
+
+
+@pytest.fixture(scope="module")
+def weights():
+    return make_clip_vit_b32_weights()
+
+
+def _torch_linear(self, x, w, b):
+    y = torch.nn.functional.linear(x.float(), w.float(), b.float() if b is not None else None)
+    return y.to(torch.bfloat16)
+
+
+def test_weights_deterministic(weights):
+    again = make_clip_vit_b32_weights()
+    for k, v in weights.items():
+        assert torch.equal(v, again[k]), k
+
+
+def test_tower_matches_transformers_fp32(weights, monkeypatch):
+    transformers = pytest.importorskip("transformers")  # noqa: F841
+    ref = oracle_vit.build_reference_clip_vision(weights)
+
+    rng = np.random.default_rng(0xC11F)
+    frames = rng.integers(0, 256, size=(2, 224, 224, 3), dtype=np.uint8)
+    pixels = clip_preprocess(frames)  # (2,3,224,224) f32
+
+    want = oracle_vit.embed_frames_fp32(ref, pixels)
+
+    monkeypatch.setattr(ClipVisionTowerAMD, "_linear", _torch_linear)
+    tower = ClipVisionTowerAMD(weights)
+    got = tower(torch.from_numpy(pixels)).float().numpy()
+
+    assert want.shape == got.shape == (2, 512)
+    cos = np.sum(want * got, axis=1)  # both unit-norm
+    assert np.all(cos >= 0.999), f"cosine too low: {cos}"
+    # unit norm
+    np.testing.assert_allclose(np.linalg.norm(got, axis=1), 1.0, atol=1e-3)
