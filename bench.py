@@ -1,0 +1,224 @@
+"""Flagship benchmark: the split/embed hot path on MI355X.
+
+Measures BASELINE.json's metric ("clips/sec + embedded frames/sec, 1080p30
+H.264") on the workload of BASELINE configs[1] — the largest single-GPU
+configuration: per clip, 21 sampled 1080p NV12 frames (10 s @ 30 fps
+sampled at 2 fps with the endpoint rule) -> fused NV12->RGB + bilinear
+resize to 224 (HIP) -> fused CLIP preprocess to bf16 NCHW (HIP) ->
+CLIP-ViT-B/32 forward with custom MFMA bf16 GEMMs -> mean-pooled
+L2-normalized clip embedding (copied to host).
+
+The timed region starts with NV12 surfaces already resident in HBM (tier
+contract; H.264 decode itself needs librocdecode, absent from this image —
+see DESIGN.md "decode roofline" for how the seam is accounted).  A step =
+one batch of --clips clips.  Multi-GPU = weak scaling: each rank processes
+its own clips; no data-path collective.
+
+Contract: one JSON line from rank 0 with metric/value/unit/... plus
+`roofline` (dominant kernel = gemm_bf16, HIP-event timed on its launch
+stream inside the timed region) and `cpu_baseline` (oracle timed on host
+cores, rank 0, N=1 only).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+FRAMES_PER_CLIP = 21  # 10 s @ 30 fps sampled at 2 fps, endpoint included
+SRC_H, SRC_W = 1088, 1920  # H.264 coded size for 1080p
+RES = 224
+# ViT-B/32 geometry for cc_gemm flop accounting (FLOPs = 2*M*N*K)
+_PATCHES, _TOKENS, _HID, _INT, _QKV, _PROJ, _LAYERS = 49, 50, 768, 3072, 2304, 512, 12
+
+
+def gemm_flops_per_frame() -> float:
+    per_layer = 2 * _TOKENS * _HID * _QKV + 2 * _TOKENS * _HID * _HID \
+        + 2 * _TOKENS * _HID * _INT + 2 * _TOKENS * _INT * _HID
+    return float(
+        2 * _PATCHES * (3 * 32 * 32) * _HID + _LAYERS * per_layer + 2 * _HID * _PROJ
+    )
+
+
+def make_nv12_batch(n_frames: int, seed: int) -> tuple[np.ndarray, np.ndarray]:
+    """Seeded moving-gradient+noise NV12 frames (BASELINE.md corpus recipe)."""
+    rng = np.random.default_rng(seed)
+    yy, xx = np.mgrid[0:SRC_H, 0:SRC_W]
+    base = ((xx * 255 // SRC_W + yy // 3) % 256).astype(np.uint8)
+    t = (np.arange(n_frames, dtype=np.int32) * 5)[:, None, None]
+    y = ((base[None].astype(np.int32) + t) % 256).astype(np.uint8)
+    noise = rng.integers(-10, 11, size=(n_frames, SRC_H // 8, SRC_W // 8), dtype=np.int16)
+    y = np.clip(
+        y.astype(np.int16) + np.kron(noise, np.ones((8, 8), dtype=np.int16)), 0, 255
+    ).astype(np.uint8)
+    uv = rng.integers(80, 176, size=(n_frames, SRC_H // 2, SRC_W), dtype=np.uint8).astype(np.uint8)
+    return y, uv
+
+
+def run_cpu_baseline(n_clips: int = 2) -> dict:
+    """Oracle (CPU restatement) on the same per-clip work, host cores."""
+    from oracle import color as ocolor
+    from oracle import vit as oracle_vit
+    from cosmos_curate_amd.models.clip_weights import make_clip_vit_b32_weights
+
+    ref = oracle_vit.build_reference_clip_vision(make_clip_vit_b32_weights())
+    y, uv = make_nv12_batch(FRAMES_PER_CLIP, seed=0)
+    t0 = time.perf_counter()
+    for _ in range(n_clips):
+        rgb = np.stack(
+            [
+                ocolor.resize_bilinear_u8(
+                    ocolor.nv12_to_rgb(y[j], uv[j].reshape(SRC_H // 2, SRC_W // 2, 2)),
+                    RES, RES,
+                )
+                for j in range(FRAMES_PER_CLIP)
+            ]
+        )
+        pix = ocolor.clip_preprocess(rgb)
+        emb = oracle_vit.embed_frames_fp32(ref, pix)
+        pooled = emb.mean(axis=0)
+        pooled /= np.linalg.norm(pooled)
+    dt = time.perf_counter() - t0
+    return {
+        "value": round(n_clips / dt, 4),
+        "unit": "clips/s",
+        "cores": torch.get_num_threads(),
+        "kind": "port",
+        "sample": f"{n_clips} clips x {FRAMES_PER_CLIP} 1080p NV12 frames, "
+        f"numpy pixel oracle + transformers fp32 ViT-B/32 ({dt:.1f}s)",
+    }
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--clips", type=int, default=16, help="clips per step per rank")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group("nccl")
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    from cosmos_curate_amd import hotpath
+    from cosmos_curate_amd.models.clip import _CLIPImageEmbeddings
+
+    lib = hotpath.require_gpu()
+    model = _CLIPImageEmbeddings()
+
+    B = args.clips
+    F = B * FRAMES_PER_CLIP
+    # inputs resident in HBM before the timed region (tier contract)
+    y_host, uv_host = make_nv12_batch(FRAMES_PER_CLIP, seed=1000 + rank)
+    y_dev = torch.from_numpy(y_host).to(device).repeat(B, 1, 1).contiguous()
+    uv_dev = torch.from_numpy(uv_host).to(device).repeat(B, 1, 1).contiguous()
+    rgb = torch.empty((F, RES, RES, 3), dtype=torch.uint8, device=device)
+    stream = torch.cuda.current_stream(device).cuda_stream
+
+    def step() -> np.ndarray:
+        hotpath.check(
+            lib.cc_nv12_to_rgb_resize(
+                y_dev.data_ptr(), uv_dev.data_ptr(), F, SRC_H, SRC_W, SRC_W,
+                rgb.data_ptr(), RES, RES, stream,
+            )
+        )
+        pixels = model.preprocess_u8(rgb)
+        emb = model.tower(pixels)  # (F, 512) f32 unit-norm
+        per_clip = emb.view(B, FRAMES_PER_CLIP, 512).mean(dim=1)
+        per_clip = per_clip / torch.linalg.vector_norm(per_clip, dim=-1, keepdim=True)
+        return per_clip.cpu().numpy()  # embeddings leave the device (16 KB)
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize(device)
+    if dist:
+        dist.barrier()
+
+    hotpath.timing_enable(True)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize(device)
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    hotpath.timing_enable(False)
+
+    gemm_ms, gemm_count = hotpath.timing_report("gemm_bf16")
+    total_clips = B * args.steps * world
+    clips_per_s = total_clips / elapsed
+    frames_per_s = clips_per_s * FRAMES_PER_CLIP
+
+    gemm_flops_step = gemm_flops_per_frame() * F
+    gemm_flops_total = gemm_flops_step * args.steps  # this rank
+    achieved = gemm_flops_total / (gemm_ms / 1e3) if gemm_ms > 0 else 0.0
+    peak = 2.5e15  # dense bf16 MFMA peak, MI355X_MICROARCH.md (spec; 2495 TF measured)
+
+    if rank == 0:
+        cpu_baseline = (
+            None
+            if (args.skip_cpu_baseline or world > 1)
+            else run_cpu_baseline()
+        )
+        out = {
+            "metric": "clips/sec + embedded frames/sec, 1080p30 H.264, 1/2/4/8 MI355X",
+            "value": round(clips_per_s, 3),
+            "unit": "clips/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no number
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "workload": "split_pipeline 1xMI355X: HIP NV12->RGB/resize + "
+                "CLIP-ViT-B/32 MFMA bf16 (decode seam excluded: no librocdecode in image)",
+                "clips_per_step": B,
+                "frames_per_clip": FRAMES_PER_CLIP,
+                "src": "1080p30 NV12 in HBM",
+                "resolution": RES,
+                "parallelism": f"dp{world}",
+                "embedded_frames_per_s": round(frames_per_s, 1),
+            },
+            "roofline": {
+                "bound": "mfma",
+                "achieved": round(achieved, 3),
+                "peak": peak,
+                "unit": "FLOP/s",
+                "frac": round(achieved / peak, 4),
+                "traffic": None,
+                "kernel": "gemm_bf16",
+                "launches": int(gemm_count),
+                "kernel_ms_total": round(gemm_ms, 3),
+            },
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -128,10 +128,21 @@ def resize_bicubic_u8(img: npt.NDArray[np.uint8], out_h: int, out_w: int) -> npt
     # gather 4 rows x 4 cols with replicate clamp
     rows = np.clip(iy[:, None] + np.arange(-1, 3)[None, :], 0, h - 1)  # (out_h,4)
     cols = np.clip(ix[:, None] + np.arange(-1, 3)[None, :], 0, w - 1)  # (out_w,4)
-    # horizontal pass: (h, out_w, C)
-    horiz = np.einsum("hwtc,wt->hwc", im[:, cols], wx.astype(np.float32))
-    # vertical pass: (out_h, out_w, C)
-    out = np.einsum("hvwc,hv->hwc", horiz[rows], wy.astype(np.float32))
+    # horizontal pass, explicit left-to-right f32 adds (same association as
+    # the HIP kernel, for bit-exact parity): ((t0+t1)+t2)+t3
+    g = im[:, cols]  # (h, out_w, 4, C)
+    wxf = wx.astype(np.float32)
+    horiz = (
+        (g[:, :, 0] * wxf[None, :, 0, None] + g[:, :, 1] * wxf[None, :, 1, None])
+        + g[:, :, 2] * wxf[None, :, 2, None]
+    ) + g[:, :, 3] * wxf[None, :, 3, None]  # (h, out_w, C) f32
+    # vertical pass, same association
+    v = horiz[rows]  # (out_h, 4, out_w, C)
+    wyf = wy.astype(np.float32)
+    out = (
+        (v[:, 0] * wyf[:, 0, None, None] + v[:, 1] * wyf[:, 1, None, None])
+        + v[:, 2] * wyf[:, 2, None, None]
+    ) + v[:, 3] * wyf[:, 3, None, None]
     return _round_u8(out.astype(np.float32))
 
 

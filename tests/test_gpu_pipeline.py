@@ -1,0 +1,137 @@
+"""End-to-end hot path on MI355X: raw-NV12 clips -> frames -> embeddings.
+
+SequentialRunner mini-pipeline (the reference's own test harness shape,
+SURVEY.md §4) over synthetic clips, with extracted frames parity-checked
+bit-exact against the oracle pixel pipeline and embeddings against the
+fp32 ViT oracle.
+"""
+
+import pathlib
+
+import numpy as np
+import pytest
+import torch
+
+from cosmos_curate_amd.core.interfaces import SequentialRunner, run_pipeline
+from cosmos_curate_amd.pipelines.video.clipping.clip_frame_extraction_stages import (
+    ClipFrameExtractionStage,
+)
+from cosmos_curate_amd.pipelines.video.embedding.clip_stages import (
+    ClipEmbeddingStage,
+    ClipFrameCreationStage,
+)
+from cosmos_curate_amd.pipelines.video.utils import raw_backend
+from cosmos_curate_amd.pipelines.video.utils.data_model import (
+    Clip,
+    SplitPipeTask,
+    Video,
+    VideoMetadata,
+)
+from oracle import color as ocolor
+from oracle import sampling as osampling
+
+pytestmark = pytest.mark.gpu
+
+FPS = 30
+SECONDS = 4  # small clip: 120 frames, sampled at 2 fps -> 9 frames
+H, W = 256, 320
+
+
+def make_task(n_clips: int = 2) -> tuple[SplitPipeTask, list[bytes]]:
+    v = Video(
+        input_video=pathlib.Path("/synthetic/clip.mp4"),
+        metadata=VideoMetadata(
+            size=1, height=H, width=W, framerate=FPS,
+            num_frames=FPS * SECONDS, duration=SECONDS, video_codec="raw",
+        ),
+    )
+    raws = []
+    import uuid
+
+    for i in range(n_clips):
+        data = raw_backend.make_synthetic_clip(FPS * SECONDS, H, W, FPS, seed=100 + i)
+        raws.append(data)
+        v.clips.append(
+            Clip(
+                uuid=uuid.uuid5(uuid.NAMESPACE_URL, f"synth_{i}"),
+                source_video="synthetic",
+                span=(0.0, float(SECONDS)),
+                encoded_data=np.frombuffer(data, dtype=np.uint8),
+            )
+        )
+    return SplitPipeTask(videos=[v]), raws
+
+
+def oracle_frames(data: bytes, fps_target: float, th: int, tw: int) -> np.ndarray:
+    ts = raw_backend.timestamps(data)
+    idx, counts, _ = osampling.sample_closest(ts, fps_target)
+    ys, uvs = raw_backend.frame_planes(data, idx)
+    n, h, w, _ = raw_backend.parse_header(data)
+    sel = np.stack(
+        [
+            ocolor.resize_bilinear_u8(
+                ocolor.nv12_to_rgb(ys[j], uvs[j].reshape(h // 2, w // 2, 2)), th, tw
+            )
+            for j in range(len(idx))
+        ]
+    )
+    return osampling.broadcast_selected(sel, np.arange(len(idx), dtype=np.int32), counts)
+
+
+def test_extraction_stage_parity():
+    task, raws = make_task(2)
+    stage = ClipFrameExtractionStage(target_fps=[2], target_res=(224, 224))
+    out = run_pipeline([task], [stage], runner=SequentialRunner())
+    video = out[0].video
+    assert not video.errors
+    for clip, raw in zip(video.clips, raws):
+        assert not clip.errors, clip.errors
+        frames = clip.extracted_frames.resolve()["FrameExtractionPolicy.sequence-2000"]
+        assert isinstance(frames, torch.Tensor) and frames.is_cuda
+        want = oracle_frames(raw, 2.0, 224, 224)
+        np.testing.assert_array_equal(frames.cpu().numpy(), want)
+
+
+def test_full_pipeline_embeddings():
+    from cosmos_curate_amd.models.clip_weights import make_clip_vit_b32_weights
+    from oracle import vit as oracle_vit
+
+    task, raws = make_task(2)
+    stages = [
+        ClipFrameExtractionStage(target_fps=[2], target_res=(224, 224)),
+        ClipFrameCreationStage(target_fps=2.0),
+        ClipEmbeddingStage(batch_size=8),
+    ]
+    out = run_pipeline([task], stages, runner=SequentialRunner())
+    video = out[0].video
+    ref = oracle_vit.build_reference_clip_vision(make_clip_vit_b32_weights())
+    for clip, raw in zip(video.clips, raws):
+        assert clip.clip_embedding is not None, clip.errors
+        assert clip.clip_embedding.shape == (512,)
+        np.testing.assert_allclose(np.linalg.norm(clip.clip_embedding), 1.0, atol=1e-3)
+        # oracle: same frames -> fp32 ViT -> mean-pool -> L2
+        frames = oracle_frames(raw, 2.0, 224, 224)
+        want = oracle_vit.embed_frames_fp32(ref, ocolor.clip_preprocess(frames))
+        pooled = want.mean(axis=0)
+        pooled /= np.linalg.norm(pooled)
+        cos = float(np.dot(pooled, clip.clip_embedding))
+        assert cos >= 0.999, cos
+
+
+def test_mp4_clip_without_rocdecode_fails_loudly():
+    """An H.264 clip on a box without librocdecode must record
+    decode_unavailable, never silently CPU-decode."""
+    task, _ = make_task(1)
+    golden = pathlib.Path(__file__).parent / "golden" / "synth_bframes.mp4"
+    task.video.clips[0].encoded_data = __import__(
+        "cosmos_curate_amd.core.utils.lazy_data", fromlist=["LazyData"]
+    ).LazyData.coerce(golden.read_bytes())
+    stage = ClipFrameExtractionStage(target_fps=[2], target_res=(224, 224))
+    out = run_pipeline([task], [stage], runner=SequentialRunner())
+    clip = out[0].video.clips[0]
+    from cosmos_curate_amd import hotpath
+
+    lib = hotpath.require_gpu()
+    if lib.cc_rocdecode_available() == 0:
+        pytest.skip("librocdecode present; decode path not yet wired")
+    assert clip.errors.get("frame_extraction") == "decode_unavailable"
