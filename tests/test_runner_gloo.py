@@ -1,0 +1,81 @@
+"""Multi-process sharding semantics on CPU (gloo, world_size=2).
+
+Covers the N>1 path of the bench/runner without GPUs: rank-sharded clip
+processing with no data-path collective, plus the
+max-over-ranks timing reduction bench.py uses.
+"""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+
+def _worker(rank: int, world: int, port: int, q) -> None:
+    try:
+        os.environ.update(
+            RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank),
+            MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+        )
+        import torch
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+
+        import pathlib
+
+        from cosmos_curate_amd.core.interfaces import CuratorStageSpec, WorkerPoolRunner
+        from cosmos_curate_amd.pipelines.video.clipping.clip_extraction_stages import (
+            FixedStrideExtractorStage,
+        )
+        from cosmos_curate_amd.pipelines.video.utils.data_model import (
+            SplitPipeTask,
+            Video,
+            VideoMetadata,
+        )
+
+        tasks = []
+        for i in range(5):
+            n = 720
+            v = Video(
+                input_video=pathlib.Path(f"/data/v{i}.mp4"),
+                metadata=VideoMetadata(size=1, height=480, width=854, framerate=24.0,
+                                       num_frames=n, duration=30.0, video_codec="h264"),
+                timestamps=(np.arange(n) / 24.0).astype(np.float32),
+            )
+            tasks.append(SplitPipeTask(videos=[v]))
+
+        runner = WorkerPoolRunner()  # reads RANK/WORLD_SIZE from env
+        out = runner.run(tasks, [CuratorStageSpec(FixedStrideExtractorStage())])
+        local_clips = sum(len(t.video.clips) for t in out)
+
+        # whole-job aggregation: sum counts, max elapsed (bench.py shape)
+        counts = torch.tensor([local_clips])
+        dist.all_reduce(counts, op=dist.ReduceOp.SUM)
+        elapsed = torch.tensor([0.1 * (rank + 1)])
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+        q.put((rank, local_clips, int(counts.item()), float(elapsed.item())))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, "ERROR", repr(e), None))
+
+
+@pytest.mark.timeout(120)
+def test_worker_pool_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=110) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+    by_rank = {r[0]: r for r in results}
+    for r in range(2):
+        assert by_rank[r][1] != "ERROR", by_rank[r]
+    # 5 videos x 3 clips = 15 total; ranks see 3/2 videos -> 9/6 clips
+    assert by_rank[0][1] == 9 and by_rank[1][1] == 6
+    assert by_rank[0][2] == by_rank[1][2] == 15  # SUM across ranks
+    assert by_rank[0][3] == pytest.approx(0.2)  # MAX over ranks
