@@ -5,25 +5,28 @@
 // QKV / attention-out projections, MLP fc1/fc2, visual projection
 // (SURVEY.md §2b row 8).
 //
-// C[M,N] = A[M,K] x B[N,K]^T (+ bias[N]); A,B bf16 row-major, accumulate
-// f32 on v_mfma_f32_16x16x32_bf16, output f32 or bf16.
+// C[M,N] = act(A[M,K] x B[N,K]^T + bias[N]) [+ residual[M,N]];
+// A,B bf16 row-major, f32 accumulate on v_mfma_f32_16x16x32_bf16, output
+// f32 or bf16.  The fused epilogue (quick-gelu, residual add) removes the
+// bandwidth-bound elementwise kernels between GEMMs (one 206 MB round
+// trip per MLP layer at the bench shape).
 //
-// Design (cdna_hip_programming.md §5 "canonical CDNA GEMM", step-3 ladder
-// structure):
-//   - 128x128 block tile, BK=64, 256 threads = 4 waves in a 2x2 wave grid,
-//     each wave owns a 64x64 sub-tile = 4x4 MFMA fragments of 16x16.
-//   - global->LDS staging via __builtin_amdgcn_global_load_lds width 16
-//     (the compiler never auto-emits it), double-buffered, one barrier +
-//     one vmcnt(0) per K-tile.
-//   - "weight layout" B[N,K] makes both operands K-major: one staging
-//     scheme, coalesced 128-byte rows, K must be a multiple of 64 (all ViT
-//     shapes are: 768/2304/3072/512).
-//   - M and N edges handled by clamping the *global* load address (garbage
-//     lands only in masked-out output rows/cols) and predicated stores.
-//
-// Numerics: f32 accumulation in k-order within each K-tile (MFMA chains);
-// parity vs torch fp32 matmul is checked at cosine/rtol level
-// (tests/test_gpu_vit.py), not bitwise — bf16 inputs round first.
+// Design (cdna_hip_programming.md §5, step-3 ladder + measured fixes):
+//   - 128x128 block tile, BK=64, 256 threads = 4 waves (2x2), each wave a
+//     64x64 sub-tile = 4x4 MFMA fragments of 16x16.
+//   - global->LDS via __builtin_amdgcn_global_load_lds width 16,
+//     double-buffered, one barrier + vmcnt(0) per K-tile.
+//   - LDS image XOR-swizzled: the 16-byte slot for (row, k16) lives at
+//     k16 ^ (row & 7).  glds demands a lane-linear LDS destination, so the
+//     swizzle is applied to the per-lane GLOBAL source address (guide §5
+//     "pre-swizzling the per-lane global address"); fragment ds_read_b128
+//     applies the same XOR.  Measured motivation: linear layout put 4
+//     same-bank rows in each 16-lane read group (SQ_LDS_BANK_CONFLICT =
+//     0.75x extra LDS cycles, profiles/r01_pmc).
+//   - XCD-aware bijective block remap (L2 affinity, guide §5.4): block b
+//     of nwg runs on XCD b%8, so adjacent output tiles share an XCD's L2.
+//   - M/N edges by clamped global loads + predicated stores; K % 64 == 0
+//     required (all ViT shapes comply).
 
 #include <hip/hip_runtime.h>
 
@@ -43,12 +46,10 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr int BM = 128, BN = 128, BK = 64;
-constexpr int WAVES_M = 2, WAVES_N = 2;  // wave grid
-constexpr int WM = BM / WAVES_M;         // 64 rows per wave
-constexpr int WN = BN / WAVES_N;         // 64 cols per wave
-constexpr int FRAG = 16;                 // mfma 16x16x32
-constexpr int MFR = WM / FRAG;           // 4 m-fragments
-constexpr int NFR = WN / FRAG;           // 4 n-fragments
+constexpr int WM = 64, WN = 64;  // per-wave sub-tile (2x2 wave grid)
+constexpr int FRAG = 16;         // mfma 16x16x32
+constexpr int MFR = WM / FRAG;   // 4
+constexpr int NFR = WN / FRAG;   // 4
 
 __device__ __forceinline__ unsigned short f32_to_bf16_rne(float v) {
   union {
@@ -59,19 +60,20 @@ __device__ __forceinline__ unsigned short f32_to_bf16_rne(float v) {
   return (unsigned short)((cv.u + 0x7fffu + lsb) >> 16);
 }
 
-// stage one 32-row slice of a [rows x BK] bf16 tile into LDS via glds.
-// lds_base: wave-uniform LDS address of this wave's 32-row slice.
-// src: global base of the matrix (bf16), ld = row stride in elements.
-// row0: first global row of the slice; nrows_clamp: clamp rows to [0, limit).
+// stage one 32-row slice of a [rows x BK] bf16 tile into LDS via glds,
+// with the k16 ^ (row&7) source swizzle.  lds_base = this wave's slice.
 __device__ __forceinline__ void stage_slice(const __bf16* __restrict__ src,
                                             long ld, long row0, long row_limit,
                                             long k0, __bf16* lds_base,
                                             int lane) {
+  const int lrow8 = lane >> 3;            // local row within each 8-row chunk
+  const int slot = lane & 7;              // LDS 16B slot this lane fills
+  const int gk16 = slot ^ lrow8;          // swizzle: global k16 feeding slot
 #pragma unroll
   for (int j = 0; j < 4; j++) {  // 4 x 1KB chunks = 32 rows
-    long grow = row0 + j * 8 + (lane >> 3);
+    long grow = row0 + j * 8 + lrow8;
     grow = grow < 0 ? 0 : (grow >= row_limit ? row_limit - 1 : grow);
-    const __bf16* gptr = src + grow * ld + k0 + (long)(lane & 7) * 8;
+    const __bf16* gptr = src + grow * ld + k0 + (long)gk16 * 8;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gptr,
         (__attribute__((address_space(3))) unsigned int*)(lds_base + j * 8 * BK),
@@ -79,13 +81,20 @@ __device__ __forceinline__ void stage_slice(const __bf16* __restrict__ src,
   }
 }
 
+// fragment read honoring the swizzle: elements (row, kk+8*(lane>>4)..+8)
+__device__ __forceinline__ bf16x8 frag_read(const __bf16* tile, int row,
+                                            int k16) {
+  int slot = k16 ^ (row & 7);
+  return *(const bf16x8*)(tile + (long)row * BK + slot * 8);
+}
+
+template <int ACT>  // 0 = none, 1 = quick-gelu
 __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
-    void* __restrict__ C, const float* __restrict__ bias, long M, long N,
-    long K, int c_is_bf16) {
-  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // As[2][128][64], Bs[2][128][64]
-  // single __shared__ object (G16 trap 4a); buffer b lives at:
-  //   A: lds + b*BM*BK            B: lds + 2*BM*BK + b*BN*BK
+    void* __restrict__ C, const float* __restrict__ bias,
+    const __bf16* __restrict__ residual, long M, long N, long K,
+    int c_is_bf16, int nbx, int nwg) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // one __shared__ object (G16 4a)
 #define AS(b) (lds + (b) * (BM * BK))
 #define BS(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
 
@@ -94,17 +103,24 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
   const int wid = tid >> 6;
   const int waveM = wid >> 1, waveN = wid & 1;
 
-  const long bm = (long)blockIdx.y * BM;
-  const long bn = (long)blockIdx.x * BN;
+  // XCD-aware bijective remap of the linear block id (guide §5.4):
+  // b ran on XCD b%8; give XCD x the contiguous tile range so its L2
+  // sees adjacent tiles.  q = nwg/8, r = nwg%8.
+  int orig = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = orig & 7, lid = orig >> 3;
+    orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+  }
+  const long bm = (long)(orig / nbx) * BM;
+  const long bn = (long)(orig % nbx) * BN;
 
-  // this wave stages rows [32*wid, 32*wid+32) of both tiles
   const long arow0 = bm + 32 * wid;
   const long brow0 = bn + 32 * wid;
 
   f32x4 acc[MFR][NFR] = {};
 
   const long KT = K / BK;
-  // prologue: stage tile 0 into buffer 0
   stage_slice(A, K, arow0, M, 0, AS(0) + 32 * wid * BK, lane);
   stage_slice(B, K, brow0, N, 0, BS(0) + 32 * wid * BK, lane);
 
@@ -117,21 +133,20 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
       stage_slice(A, K, arow0, M, k0, AS(buf ^ 1) + 32 * wid * BK, lane);
       stage_slice(B, K, brow0, N, k0, BS(buf ^ 1) + 32 * wid * BK, lane);
     }
-    // compute on buf: 2 k-steps of 32, 16 MFMA each
     const __bf16* At = AS(buf);
     const __bf16* Bt = BS(buf);
     const int arow_frag = waveM * WM + (lane & 15);
     const int brow_frag = waveN * WN + (lane & 15);
-    const int koff = 8 * (lane >> 4);
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 32) {
+      const int k16 = (kk >> 3) + (lane >> 4);
       bf16x8 afrag[MFR], bfrag[NFR];
 #pragma unroll
       for (int m = 0; m < MFR; m++)
-        afrag[m] = *(const bf16x8*)(At + (arow_frag + m * FRAG) * BK + kk + koff);
+        afrag[m] = frag_read(At, arow_frag + m * FRAG, k16);
 #pragma unroll
       for (int n = 0; n < NFR; n++)
-        bfrag[n] = *(const bf16x8*)(Bt + (brow_frag + n * FRAG) * BK + kk + koff);
+        bfrag[n] = frag_read(Bt, brow_frag + n * FRAG, k16);
 #pragma unroll
       for (int m = 0; m < MFR; m++)
 #pragma unroll
@@ -157,6 +172,8 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
         const long row = crow_base + m * FRAG + r;
         if (row >= M) continue;
         float v = acc[m][n][r] + bval;
+        if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));  // quick-gelu
+        if (residual) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
         else
@@ -176,17 +193,21 @@ inline void record_timing(const char* name, float ms) {
 
 }  // namespace
 
-extern "C" int cc_gemm_bf16(const void* A, const void* B, void* C, int64_t M,
-                            int64_t N, int64_t K, const float* bias,
-                            int c_dtype, uint64_t stream) {
+extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
+                               int64_t M, int64_t N, int64_t K,
+                               const float* bias, int c_dtype, int act,
+                               const void* residual, uint64_t stream) {
   if (!A || !B || !C || M <= 0 || N <= 0 || K <= 0)
     return cc::set_error(CC_ERR_INVALID, "bad gemm args");
   if (K % BK != 0)
     return cc::set_error(CC_ERR_UNSUPPORTED,
                          "cc_gemm_bf16 requires K %% 64 == 0 (got %lld)",
                          (long long)K);
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
   dim3 block(256);
-  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+  dim3 grid(nwg);
   auto& ts = cc::timing();
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
   bool timed = false;
@@ -196,9 +217,16 @@ extern "C" int cc_gemm_bf16(const void* A, const void* B, void* C, int64_t M,
       timed = true;
     }
   }
-  hipLaunchKernelGGL(k_gemm_bf16, grid, block, 0, (hipStream_t)stream,
-                     (const __bf16*)A, (const __bf16*)B, C, bias, (long)M,
-                     (long)N, (long)K, c_dtype == 1 ? 1 : 0);
+  if (act == 1)
+    hipLaunchKernelGGL(k_gemm_bf16<1>, grid, block, 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C, bias,
+                       (const __bf16*)residual, (long)M, (long)N, (long)K,
+                       c_dtype == 1 ? 1 : 0, nbx, nwg);
+  else
+    hipLaunchKernelGGL(k_gemm_bf16<0>, grid, block, 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C, bias,
+                       (const __bf16*)residual, (long)M, (long)N, (long)K,
+                       c_dtype == 1 ? 1 : 0, nbx, nwg);
   hipError_t e = hipGetLastError();
   if (timed) {
     hipEventRecord(ev1, (hipStream_t)stream);
@@ -212,4 +240,10 @@ extern "C" int cc_gemm_bf16(const void* A, const void* B, void* C, int64_t M,
   if (e != hipSuccess)
     return cc::set_error(CC_ERR_HIP, "gemm launch: %s", hipGetErrorString(e));
   return CC_OK;
+}
+
+extern "C" int cc_gemm_bf16(const void* A, const void* B, void* C, int64_t M,
+                            int64_t N, int64_t K, const float* bias,
+                            int c_dtype, uint64_t stream) {
+  return cc_gemm_bf16_ex(A, B, C, M, N, K, bias, c_dtype, 0, nullptr, stream);
 }

@@ -63,6 +63,9 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.cc_gemm_bf16.argtypes = [
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64, c.c_int64,
         c.c_void_p, c.c_int, c.c_uint64]
+    lib.cc_gemm_bf16_ex.argtypes = [
+        c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64, c.c_int64,
+        c.c_void_p, c.c_int, c.c_int, c.c_void_p, c.c_uint64]
 
     lib.cc_timing_enable.argtypes = [c.c_int]
     lib.cc_timing_report.argtypes = [

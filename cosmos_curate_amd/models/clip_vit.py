@@ -29,8 +29,16 @@ from cosmos_curate_amd import hotpath
 from cosmos_curate_amd.models import clip_weights as cw
 
 
-def _cc_linear(x: torch.Tensor, w_bf16: torch.Tensor, bias_f32: torch.Tensor | None) -> torch.Tensor:
-    """C[M,N] = x[M,K] @ w[N,K]^T + bias via cc_gemm_bf16 (bf16 out)."""
+def _cc_linear(
+    x: torch.Tensor,
+    w_bf16: torch.Tensor,
+    bias_f32: torch.Tensor | None,
+    act: int = 0,
+    residual: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """C[M,N] = act(x[M,K] @ w[N,K]^T + bias) [+ residual], bf16 out.
+
+    act 1 = quick-gelu; residual fused into the epilogue (cc_gemm_bf16_ex)."""
     lib = hotpath.require_gpu()
     assert x.dtype == torch.bfloat16 and w_bf16.dtype == torch.bfloat16
     x = x.contiguous()
@@ -38,11 +46,15 @@ def _cc_linear(x: torch.Tensor, w_bf16: torch.Tensor, bias_f32: torch.Tensor | N
     N = w_bf16.shape[0]
     out = torch.empty((M, N), dtype=torch.bfloat16, device=x.device)
     stream = torch.cuda.current_stream(x.device).cuda_stream
+    if residual is not None:
+        residual = residual.contiguous()
+        assert residual.shape == out.shape and residual.dtype == torch.bfloat16
     hotpath.check(
-        lib.cc_gemm_bf16(
+        lib.cc_gemm_bf16_ex(
             x.data_ptr(), w_bf16.data_ptr(), out.data_ptr(), M, N, K,
             bias_f32.data_ptr() if bias_f32 is not None else None,
-            1, stream,
+            1, act, residual.data_ptr() if residual is not None else None,
+            stream,
         )
     )
     return out
@@ -91,8 +103,15 @@ class ClipVisionTowerAMD(torch.nn.Module):
         self.scale = 1.0 / math.sqrt(cw.HIDDEN // cw.HEADS)
 
     # the one contraction primitive; tests may monkeypatch this
-    def _linear(self, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None) -> torch.Tensor:
-        return _cc_linear(x, w, b)
+    def _linear(
+        self,
+        x: torch.Tensor,
+        w: torch.Tensor,
+        b: torch.Tensor | None,
+        act: int = 0,
+        residual: torch.Tensor | None = None,
+    ) -> torch.Tensor:
+        return _cc_linear(x, w, b, act, residual)
 
     def _ln(self, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
         return torch.nn.functional.layer_norm(
@@ -134,22 +153,25 @@ class ClipVisionTowerAMD(torch.nn.Module):
                 q, k, v, scale=self.scale
             )
             attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cw.HIDDEN)
-            h = res + self._linear(
-                attn, getattr(self, f"w_out_{i}"), getattr(self, f"b_out_{i}")
+            # residual add fused into the out-proj epilogue
+            h = self._linear(
+                attn, getattr(self, f"w_out_{i}"), getattr(self, f"b_out_{i}"),
+                residual=res.reshape(n * seq, cw.HIDDEN),
             ).reshape(n, seq, cw.HIDDEN)
 
             res = h
             y = self._ln(h, getattr(self, f"ln2_w_{i}"), getattr(self, f"ln2_b_{i}"))
+            # quick-gelu fused into the fc1 epilogue (transformers CLIP act)
             y = self._linear(
                 y.reshape(n * seq, cw.HIDDEN),
                 getattr(self, f"w_fc1_{i}"),
                 getattr(self, f"b_fc1_{i}"),
+                act=1,
             )
-            y = y * torch.sigmoid(1.702 * y)  # quick_gelu (transformers CLIP)
-            y = self._linear(
-                y, getattr(self, f"w_fc2_{i}"), getattr(self, f"b_fc2_{i}")
+            h = self._linear(
+                y, getattr(self, f"w_fc2_{i}"), getattr(self, f"b_fc2_{i}"),
+                residual=res.reshape(n * seq, cw.HIDDEN),
             ).reshape(n, seq, cw.HIDDEN)
-            h = res + y
 
         pooled = self._ln(h[:, 0], self.post_ln_w, self.post_ln_b)
         emb = self._linear(pooled.to(torch.bfloat16), self.w_proj, None).float()

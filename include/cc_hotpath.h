@@ -139,6 +139,14 @@ int cc_gather_frames_u8(const void* frames, int n_in, size_t frame_bytes,
 int cc_gemm_bf16(const void* A, const void* B, void* C,
                  int64_t M, int64_t N, int64_t K,
                  const float* bias, int c_dtype, uint64_t stream);
+/* Fused-epilogue form: act 0 = none, 1 = quick-gelu (x*sigmoid(1.702x),
+ * the CLIP MLP activation); residual (bf16 [M,N], may be NULL) is added
+ * after the activation — fuses the MLP/attention elementwise kernels into
+ * the producing GEMM. */
+int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
+                    int64_t M, int64_t N, int64_t K,
+                    const float* bias, int c_dtype, int act,
+                    const void* residual, uint64_t stream);
 
 /* ---- kernel timing (bench.py roofline evidence) ---------------------
  * When enabled, every cc_* kernel launch is bracketed with hipEvents on

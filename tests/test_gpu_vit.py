@@ -69,6 +69,25 @@ def test_gemm_bf16_out_and_no_bias(lib):
     torch.testing.assert_close(got, want, rtol=2e-2, atol=5e-2)
 
 
+def test_gemm_fused_gelu_and_residual(lib):
+    torch.manual_seed(3)
+    M, N, K = 200, 256, 128
+    a = torch.randn(M, K).to(torch.bfloat16).cuda()
+    b = torch.randn(N, K).to(torch.bfloat16).cuda()
+    bias = torch.randn(N).float().cuda()
+    res = torch.randn(M, N).to(torch.bfloat16).cuda()
+    out = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(
+        lib.cc_gemm_bf16_ex(a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K,
+                            bias.data_ptr(), 1, 1, res.data_ptr(), stream)
+    )
+    torch.cuda.synchronize()
+    y = a.float().cpu() @ b.float().cpu().T + bias.cpu()
+    y = y * torch.sigmoid(1.702 * y) + res.float().cpu()
+    torch.testing.assert_close(out.float().cpu(), y, rtol=2e-2, atol=8e-2)
+
+
 def test_gemm_rejects_bad_k(lib):
     a = torch.randn(16, 60).to(torch.bfloat16).cuda()
     b = torch.randn(16, 60).to(torch.bfloat16).cuda()

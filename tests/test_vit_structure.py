@@ -24,8 +24,12 @@ def weights():
     return make_clip_vit_b32_weights()
 
 
-def _torch_linear(self, x, w, b):
+def _torch_linear(self, x, w, b, act=0, residual=None):
     y = torch.nn.functional.linear(x.float(), w.float(), b.float() if b is not None else None)
+    if act == 1:
+        y = y * torch.sigmoid(1.702 * y)
+    if residual is not None:
+        y = y + residual.float()
     return y.to(torch.bfloat16)
 
 
