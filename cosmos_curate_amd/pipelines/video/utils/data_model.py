@@ -218,10 +218,37 @@ class Video:
 
 @dataclasses.dataclass
 class SplitPipeTask(PipelineTask):
-    """data_model.py:690-802 hot-path subset (single-cam)."""
+    """data_model.py:690-802 subset.
 
+    session_id = the video path for single-camera tasks (data_model.py:695);
+    ``video=`` kwarg accepted for single-cam construction (:710-727).
+    """
+
+    session_id: str = ""
     videos: list[Video] = dataclasses.field(default_factory=list)
     stage_perf: dict[str, Any] = dataclasses.field(default_factory=dict)
+    errors: dict[str, str] = dataclasses.field(default_factory=dict)
+    _init_video: Video | None = dataclasses.field(default=None, repr=False)
+
+    def __init__(
+        self,
+        session_id: str = "",
+        videos: list[Video] | None = None,
+        stage_perf: dict[str, Any] | None = None,
+        errors: dict[str, str] | None = None,
+        video: Video | None = None,
+    ) -> None:
+        if video is not None:
+            if videos:
+                msg = "Cannot specify both 'video' and 'videos' parameters"
+                raise ValueError(msg)
+            videos = [video]
+        self.session_id = session_id
+        self.videos = videos if videos is not None else []
+        self.stage_perf = stage_perf if stage_perf is not None else {}
+        self.errors = errors if errors is not None else {}
+        if not self.session_id and self.videos:
+            self.session_id = self.videos[0].input_path
 
     @property
     def video(self) -> Video:
