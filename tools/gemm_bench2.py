@@ -1,0 +1,101 @@
+"""Compare GEMM structure variants (tools/gemm_variants.hip) vs the
+production kernel on the ViT shapes.  Builds the variants .so on the fly.
+
+    gpurun -- 'python tools/gemm_bench2.py'
+"""
+
+from __future__ import annotations
+
+import ctypes
+import pathlib
+import subprocess
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+
+from cosmos_curate_amd import hotpath  # noqa: E402
+
+ROOT = pathlib.Path(__file__).resolve().parent.parent
+SO = ROOT / "tools" / "libgemm_variants.so"
+
+SHAPES = [
+    ("patch", 336 * 49, 768, 3072),
+    ("qkv", 336 * 50, 2304, 768),
+    ("attn_out", 336 * 50, 768, 768),
+    ("fc1", 336 * 50, 3072, 768),
+    ("fc2", 336 * 50, 768, 3072),
+    ("square4k", 4096, 4096, 4096),
+    ("square8k", 8192, 8192, 8192),
+]
+
+
+def build() -> ctypes.CDLL:
+    src = ROOT / "tools" / "gemm_variants.hip"
+    if not SO.exists() or SO.stat().st_mtime < src.stat().st_mtime:
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+             f"-I{ROOT}", "-shared", str(src), "-o", str(SO)],
+            check=True,
+        )
+    lib = ctypes.CDLL(str(SO))
+    lib.cc_gemm_variant.argtypes = [
+        ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.c_int64, ctypes.c_int64, ctypes.c_int, ctypes.c_uint64,
+    ]
+    return lib
+
+
+def time_variant(fn, iters=30) -> float:
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main() -> None:
+    prod = hotpath.require_gpu()
+    var = build()
+    stream = torch.cuda.current_stream().cuda_stream
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2(32x32)':>9s} {'v3(3buf)':>8s}  TF/s")
+    for label, M, N, K in SHAPES:
+        torch.manual_seed(1)
+        a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
+        b = (torch.randn(N, K) * 0.3).to(torch.bfloat16).cuda()
+        c = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+        want = None
+
+        def prod_call():
+            hotpath.check(prod.cc_gemm_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, None, 1, stream))
+
+        results = []
+        prod_call()
+        torch.cuda.synchronize()
+        want = c[:128, :128].float().cpu().clone()
+        dt = time_variant(prod_call, 20 if M * K > 2**24 else 30)
+        results.append(2.0 * M * N * K / dt / 1e12)
+        for v in (2, 3):
+            def vcall(v=v):
+                rc = var.cc_gemm_variant(v, a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
+                assert rc == 0
+            c.zero_()
+            vcall()
+            torch.cuda.synchronize()
+            got = c[:128, :128].float().cpu()
+            err = (got - want).abs().max().item()
+            ok = err < 0.2
+            if not ok:
+                results.append(float("nan"))
+                print(f"  variant {v} WRONG on {label}: max err {err}")
+                continue
+            dt = time_variant(vcall, 20 if M * K > 2**24 else 30)
+            results.append(2.0 * M * N * K / dt / 1e12)
+        print(f"{label:9s} {M:6d} {N:5d} {K:5d} | {results[0]:7.1f} {results[1]:9.1f} {results[2]:8.1f}")
+
+
+if __name__ == "__main__":
+    main()
