@@ -61,40 +61,45 @@ def main() -> None:
     prod = hotpath.require_gpu()
     var = build()
     stream = torch.cuda.current_stream().cuda_stream
-    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2(32x32)':>9s} {'v3(3buf)':>8s}  TF/s")
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v3':>7s} {'v4':>7s}  TF/s (best of 3 reps)")
     for label, M, N, K in SHAPES:
         torch.manual_seed(1)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
         b = (torch.randn(N, K) * 0.3).to(torch.bfloat16).cuda()
         c = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
-        want = None
 
         def prod_call():
             hotpath.check(prod.cc_gemm_bf16(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, None, 1, stream))
 
-        results = []
         prod_call()
         torch.cuda.synchronize()
         want = c[:128, :128].float().cpu().clone()
-        dt = time_variant(prod_call, 20 if M * K > 2**24 else 30)
-        results.append(2.0 * M * N * K / dt / 1e12)
-        for v in (2, 3):
+        iters = 20 if M * N * K > 2**36 else 40
+        results = []
+        calls = {0: prod_call}
+        for v in (2, 3, 4):
             def vcall(v=v):
                 rc = var.cc_gemm_variant(v, a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
                 assert rc == 0
-            c.zero_()
-            vcall()
-            torch.cuda.synchronize()
-            got = c[:128, :128].float().cpu()
-            err = (got - want).abs().max().item()
-            ok = err < 0.2
-            if not ok:
-                results.append(float("nan"))
-                print(f"  variant {v} WRONG on {label}: max err {err}")
-                continue
-            dt = time_variant(vcall, 20 if M * K > 2**24 else 30)
-            results.append(2.0 * M * N * K / dt / 1e12)
-        print(f"{label:9s} {M:6d} {N:5d} {K:5d} | {results[0]:7.1f} {results[1]:9.1f} {results[2]:8.1f}")
+            calls[v] = vcall
+        for v, fn in calls.items():
+            if v != 0:
+                c.zero_()
+                fn()
+                torch.cuda.synchronize()
+                err = (c[:128, :128].float().cpu() - want).abs().max().item()
+                if err > 0.2:
+                    results.append(float("nan"))
+                    print(f"  variant {v} WRONG on {label}: max err {err}")
+                    continue
+            for _ in range(5):  # warmup per variant
+                fn()
+            best = min(time_variant(fn, iters) for _ in range(3))
+            results.append(2.0 * M * N * K / best / 1e12)
+        print(
+            f"{label:9s} {M:6d} {N:5d} {K:5d} | "
+            + " ".join(f"{r:7.1f}" for r in results)
+        )
 
 
 if __name__ == "__main__":

@@ -200,6 +200,79 @@ __global__ __launch_bounds__(256, 2) void k_gemm_v3(
 #undef BS3
 }
 
+// ---- V4: 32x32x16 fragments + 3-buffer counted vmcnt ----
+__global__ __launch_bounds__(256, 2) void k_gemm_v4(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16) {
+  __shared__ __bf16 lds[3 * (BM + BN) * BK];
+#define AS4(b) (lds + (b) * (BM * BK))
+#define BS4(b) (lds + 3 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int waveM = wid >> 1, waveN = wid & 1;
+  const long bm = (long)blockIdx.y * BM, bn = (long)blockIdx.x * BN;
+  const long arow0 = bm + 32 * wid, brow0 = bn + 32 * wid;
+
+  f32x16 acc[2][2] = {};
+  const long KT = K / BK;
+  stage_slice(A, K, arow0, M, 0, AS4(0) + 32 * wid * BK, lane);
+  stage_slice(B, K, brow0, N, 0, BS4(0) + 32 * wid * BK, lane);
+  if (KT > 1) {
+    stage_slice(A, K, arow0, M, BK, AS4(1) + 32 * wid * BK, lane);
+    stage_slice(B, K, brow0, N, BK, BS4(1) + 32 * wid * BK, lane);
+  }
+  for (long kt = 0; kt < KT; ++kt) {
+    const int buf = kt % 3;
+    if (kt + 1 < KT)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (kt + 2 < KT) {
+      const long k0 = (kt + 2) * BK;
+      const int nb = (kt + 2) % 3;
+      stage_slice(A, K, arow0, M, k0, AS4(nb) + 32 * wid * BK, lane);
+      stage_slice(B, K, brow0, N, k0, BS4(nb) + 32 * wid * BK, lane);
+    }
+    const __bf16* At = AS4(buf);
+    const __bf16* Bt = BS4(buf);
+    const int arow = waveM * 64 + (lane & 31);
+    const int brow = waveN * 64 + (lane & 31);
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      const int k16 = (kk >> 3) + (lane >> 5);
+      bf16x8 a0 = frag_read(At, arow, k16);
+      bf16x8 a1 = frag_read(At, arow + 32, k16);
+      bf16x8 b0 = frag_read(Bt, brow, k16);
+      bf16x8 b1 = frag_read(Bt, brow + 32, k16);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+  }
+  const long col0 = bn + waveN * 64 + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 2; m++)
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const long col = col0 + n * 32;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const long row = bm + waveM * 64 + m * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        if (row >= M) continue;
+        float v = acc[m][n][reg];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+#undef AS4
+#undef BS4
+}
+
 }  // namespace
 
 extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* C,
@@ -216,8 +289,12 @@ extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* 
     hipLaunchKernelGGL(k_gemm_v3, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, (long)M, (long)N,
                        (long)K, c_dtype == 1);
+  else if (variant == 4)
+    hipLaunchKernelGGL(k_gemm_v4, grid, block, 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C, (long)M, (long)N,
+                       (long)K, c_dtype == 1);
   else
-    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3");
+    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4");
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return cc::set_error(CC_ERR_HIP, "%s", hipGetErrorString(e));
   return CC_OK;
