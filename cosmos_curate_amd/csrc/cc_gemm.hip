@@ -44,6 +44,7 @@ namespace {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 constexpr int BM = 128, BN = 128, BK = 64;
 constexpr int WM = 64, WN = 64;  // per-wave sub-tile (2x2 wave grid)
@@ -183,6 +184,96 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
   }
 }
 
+// 32x32x16-fragment variant: same staging/swizzle/XCD remap, wave tile
+// 64x64 as 2x2 fragments of 32x32.  Measured (profiles/r01_gemm_variants):
+// +19-36% over the 16x16x32 body on the K=768 ViT shapes and +7% on big
+// squares (fewer, denser MFMAs: 2382 vs 2075 TF ceiling); the 16x16 body
+// keeps a small edge at K >= 2048 (patch-embed), so the launcher picks by K.
+template <int ACT>
+__global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    const __bf16* __restrict__ residual, long M, long N, long K,
+    int c_is_bf16, int nbx, int nwg) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];
+#define AS32(b) (lds + (b) * (BM * BK))
+#define BS32(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 1, waveN = wid & 1;
+
+  int orig = blockIdx.x;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = orig & 7, lid = orig >> 3;
+    orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+  }
+  const long bm = (long)(orig / nbx) * BM;
+  const long bn = (long)(orig % nbx) * BN;
+  const long arow0 = bm + 32 * wid;
+  const long brow0 = bn + 32 * wid;
+
+  f32x16 acc[2][2] = {};
+  const long KT = K / BK;
+  stage_slice(A, K, arow0, M, 0, AS32(0) + 32 * wid * BK, lane);
+  stage_slice(B, K, brow0, N, 0, BS32(0) + 32 * wid * BK, lane);
+  int buf = 0;
+  for (long kt = 0; kt < KT; ++kt) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (kt + 1 < KT) {
+      const long k0 = (kt + 1) * BK;
+      stage_slice(A, K, arow0, M, k0, AS32(buf ^ 1) + 32 * wid * BK, lane);
+      stage_slice(B, K, brow0, N, k0, BS32(buf ^ 1) + 32 * wid * BK, lane);
+    }
+    const __bf16* At = AS32(buf);
+    const __bf16* Bt = BS32(buf);
+    const int arow = waveM * WM + (lane & 31);
+    const int brow = waveN * WN + (lane & 31);
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      // 32x32x16 A/B fragment: lane l holds row l&31, k 8*(l>>5)..+8
+      const int k16 = (kk >> 3) + (lane >> 5);
+      bf16x8 a0 = frag_read(At, arow, k16);
+      bf16x8 a1 = frag_read(At, arow + 32, k16);
+      bf16x8 b0 = frag_read(Bt, brow, k16);
+      bf16x8 b1 = frag_read(Bt, brow + 32, k16);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    buf ^= 1;
+  }
+  // C/D map 32x32x16: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  const long col0 = bn + waveN * WN + (lane & 31);
+#pragma unroll
+  for (int m = 0; m < 2; m++) {
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const long col = col0 + n * 32;
+      if (col >= N) continue;
+      const float bval = bias ? bias[col] : 0.0f;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const long row =
+            bm + waveM * WM + m * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        if (row >= M) continue;
+        float v = acc[m][n][reg] + bval;
+        if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+        if (residual) v += (float)residual[row * N + col];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+  }
+#undef AS32
+#undef BS32
+}
+
 inline void record_timing(const char* name, float ms) {
   auto& ts = cc::timing();
   std::lock_guard<std::mutex> lk(ts.mu);
@@ -217,7 +308,21 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
       timed = true;
     }
   }
-  if (act == 1)
+  // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
+  // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
+  const bool wide = K < 2048;
+  if (wide) {
+    if (act == 1)
+      hipLaunchKernelGGL(k_gemm_bf16_w32<1>, grid, block, 0, (hipStream_t)stream,
+                         (const __bf16*)A, (const __bf16*)B, C, bias,
+                         (const __bf16*)residual, (long)M, (long)N, (long)K,
+                         c_dtype == 1 ? 1 : 0, nbx, nwg);
+    else
+      hipLaunchKernelGGL(k_gemm_bf16_w32<0>, grid, block, 0, (hipStream_t)stream,
+                         (const __bf16*)A, (const __bf16*)B, C, bias,
+                         (const __bf16*)residual, (long)M, (long)N, (long)K,
+                         c_dtype == 1 ? 1 : 0, nbx, nwg);
+  } else if (act == 1)
     hipLaunchKernelGGL(k_gemm_bf16<1>, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, bias,
                        (const __bf16*)residual, (long)M, (long)N, (long)K,

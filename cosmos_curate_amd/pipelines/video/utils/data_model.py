@@ -93,6 +93,8 @@ class Clip:
     # embedding produced by the configured embedder (CLIP/SigLIP)
     clip_embedding: npt.NDArray[np.float32] | None = None
     clip_embedding_frames: LazyData = dataclasses.field(default_factory=LazyData)
+    # aesthetic filter (data_model.py:241)
+    aesthetic_score: float | None = None
     errors: dict[str, str] = dataclasses.field(default_factory=dict)
 
     def __post_init__(self) -> None:
