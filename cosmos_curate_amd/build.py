@@ -30,6 +30,7 @@ SOURCES: list[tuple[str, list[str]]] = [
     ("cc_decode.cpp", []),
     ("cc_pixel.hip", [ARCH, "-ffp-contract=off"]),
     ("cc_gemm.hip", [ARCH]),
+    ("cc_dedup.hip", [ARCH]),
 ]
 
 

@@ -26,15 +26,25 @@ inline int set_error(int code, const char* fmt, ...) {
 }
 
 // ---- kernel timing registry (hipEvent based, see cc_timing_* ABI) ----
+// Launches are bracketed with event pairs on their stream but NOT
+// synchronized at launch time (that would serialize the pipeline and
+// time isolated launches); pending pairs drain at cc_timing_report.
 struct TimingEntry {
   double total_ms = 0.0;
   int64_t count = 0;
+};
+
+struct PendingPair {
+  std::string name;
+  void* ev0;
+  void* ev1;
 };
 
 struct TimingState {
   std::mutex mu;
   bool enabled = false;
   std::map<std::string, TimingEntry> entries;
+  std::vector<PendingPair> pending;
 };
 
 inline TimingState& timing() {

@@ -30,7 +30,10 @@
 
 #include <hip/hip_runtime.h>
 
+#include <vector>
+
 #include "cc_common.hpp"
+#include "cc_timing.hpp"
 
 #define CC_CHECK_HIP(expr)                                                   \
   do {                                                                       \
@@ -274,14 +277,6 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
 #undef BS32
 }
 
-inline void record_timing(const char* name, float ms) {
-  auto& ts = cc::timing();
-  std::lock_guard<std::mutex> lk(ts.mu);
-  auto& e = ts.entries[name];
-  e.total_ms += ms;
-  e.count += 1;
-}
-
 }  // namespace
 
 extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
@@ -299,15 +294,8 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   int nwg = nbx * nby;
   dim3 block(256);
   dim3 grid(nwg);
-  auto& ts = cc::timing();
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
-  bool timed = false;
-  if (ts.enabled) {
-    if (hipEventCreate(&ev0) == hipSuccess && hipEventCreate(&ev1) == hipSuccess) {
-      hipEventRecord(ev0, (hipStream_t)stream);
-      timed = true;
-    }
-  }
+  bool timed = cc::timed_begin(stream, &ev0, &ev1);
   // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
   // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
   const bool wide = K < 2048;
@@ -333,15 +321,7 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
                        (const __bf16*)residual, (long)M, (long)N, (long)K,
                        c_dtype == 1 ? 1 : 0, nbx, nwg);
   hipError_t e = hipGetLastError();
-  if (timed) {
-    hipEventRecord(ev1, (hipStream_t)stream);
-    hipEventSynchronize(ev1);
-    float ms = 0;
-    hipEventElapsedTime(&ms, ev0, ev1);
-    record_timing("gemm_bf16", ms);
-    hipEventDestroy(ev0);
-    hipEventDestroy(ev1);
-  }
+  if (timed) cc::timed_end("gemm_bf16", stream, ev0, ev1);
   if (e != hipSuccess)
     return cc::set_error(CC_ERR_HIP, "gemm launch: %s", hipGetErrorString(e));
   return CC_OK;

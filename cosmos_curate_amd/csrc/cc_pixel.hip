@@ -19,7 +19,10 @@
 
 #include <hip/hip_runtime.h>
 
+#include <vector>
+
 #include "cc_common.hpp"
+#include "cc_timing.hpp"
 
 #define CC_CHECK_HIP(expr)                                                   \
   do {                                                                       \
@@ -296,36 +299,18 @@ __global__ void k_gather_frames_u8(const unsigned char* __restrict__ frames,
   out[idx] = frames[(size_t)out_src[j] * frame_bytes + b];
 }
 
-// timing helper
+// timing helpers (non-serializing; see cc_timing.hpp)
 inline int launch_timed(const char* name, uint64_t stream, hipEvent_t* evs,
                         bool* timed) {
-  auto& ts = cc::timing();
-  *timed = false;
-  if (ts.enabled) {
-    if (hipEventCreate(&evs[0]) != hipSuccess) return 0;
-    if (hipEventCreate(&evs[1]) != hipSuccess) return 0;
-    hipEventRecord(evs[0], (hipStream_t)stream);
-    *timed = true;
-  }
+  (void)name;
+  *timed = cc::timed_begin(stream, &evs[0], &evs[1]);
   return 0;
 }
 
 inline void finish_timed(const char* name, uint64_t stream, hipEvent_t* evs,
                          bool timed) {
   if (!timed) return;
-  auto& ts = cc::timing();
-  hipEventRecord(evs[1], (hipStream_t)stream);
-  hipEventSynchronize(evs[1]);
-  float ms = 0;
-  hipEventElapsedTime(&ms, evs[0], evs[1]);
-  {
-    std::lock_guard<std::mutex> lk(ts.mu);
-    auto& e = ts.entries[name];
-    e.total_ms += ms;
-    e.count += 1;
-  }
-  hipEventDestroy(evs[0]);
-  hipEventDestroy(evs[1]);
+  cc::timed_end(name, stream, evs[0], evs[1]);
 }
 
 }  // namespace
@@ -369,12 +354,14 @@ int cc_timing_enable(int enable) {
   return CC_OK;
 }
 int cc_timing_reset(void) {
+  cc::timed_drain();  // release any pending events
   auto& ts = cc::timing();
   std::lock_guard<std::mutex> lk(ts.mu);
   ts.entries.clear();
   return CC_OK;
 }
 int cc_timing_report(const char* kernel, double* total_ms, int64_t* count) {
+  cc::timed_drain();
   auto& ts = cc::timing();
   std::lock_guard<std::mutex> lk(ts.mu);
   auto it = ts.entries.find(kernel);

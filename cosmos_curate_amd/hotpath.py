@@ -67,6 +67,9 @@ def _declare(lib: ctypes.CDLL) -> None:
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64, c.c_int64,
         c.c_void_p, c.c_int, c.c_int, c.c_void_p, c.c_uint64]
 
+    lib.cc_pairwise_max_earlier.argtypes = [
+        c.c_void_p, c.c_int64, c.c_int64, c.c_void_p, c.c_void_p, c.c_uint64]
+
     lib.cc_timing_enable.argtypes = [c.c_int]
     lib.cc_timing_report.argtypes = [
         c.c_char_p, c.POINTER(c.c_double), c.POINTER(c.c_int64)]

@@ -148,6 +148,14 @@ int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
                     const float* bias, int c_dtype, int act,
                     const void* residual, uint64_t stream);
 
+/* ---- semantic dedup -------------------------------------------------
+ * Strict-upper-triangular max-cosine scan (SemDedupActor.dedup,
+ * pipelines/video/dedup/dedup_actor.py:315-460): for each row j of the
+ * L2-normalized f32 matrix e[m][d] (scan order), the max cosine to any
+ * earlier row and its argmax.  Row 0 -> (0.0, 0).  d % 64 == 0. */
+int cc_pairwise_max_earlier(const void* e_f32, int64_t m, int64_t d,
+                            void* maxv_f32, void* argi_i32, uint64_t stream);
+
 /* ---- kernel timing (bench.py roofline evidence) ---------------------
  * When enabled, every cc_* kernel launch is bracketed with hipEvents on
  * its launch stream; totals are accumulated per kernel name. */
