@@ -103,15 +103,27 @@ def _assemble_stages(args: argparse.Namespace) -> list[CuratorStage | CuratorSta
     stages.append(
         ClipTranscodingStage(num_clips_per_chunk=args.num_clips_per_chunk, log_stats=True)
     )
-    if args.generate_embeddings:
+    if args.generate_embeddings or args.aesthetic_threshold is not None:
+        fps_targets: list[float | int] = [args.target_clip_fps]
+        if args.aesthetic_threshold is not None and 1.0 not in fps_targets:
+            fps_targets.append(1.0)  # aesthetics samples at 1 fps (builder default)
         stages.append(
             ClipFrameExtractionStage(
                 extraction_policies=(FrameExtractionPolicy.sequence,),
-                target_fps=[args.target_clip_fps],
+                target_fps=sorted(fps_targets),
                 target_res=(224, 224),
                 log_stats=True,
             )
         )
+    if args.aesthetic_threshold is not None:
+        from cosmos_curate_amd.pipelines.video.filtering.aesthetics.aesthetic_filter_stages import (
+            AestheticFilterStage,
+        )
+
+        stages.append(
+            AestheticFilterStage(score_threshold=args.aesthetic_threshold, log_stats=True)
+        )
+    if args.generate_embeddings:
         stages.append(ClipFrameCreationStage(target_fps=args.target_clip_fps, log_stats=True))
         stages.append(ClipEmbeddingStage(log_stats=True))
     stages.append(ClipWriterStage(args.output_clip_path, log_stats=True))
@@ -136,6 +148,7 @@ def _setup_parser(parser: argparse.ArgumentParser) -> None:
     parser.add_argument("--generate-embeddings", action="store_true", default=True)
     parser.add_argument("--no-embeddings", dest="generate_embeddings", action="store_false")
     parser.add_argument("--embedding-algorithm", default="clip")
+    parser.add_argument("--aesthetic-threshold", type=float, default=None)
     parser.add_argument("--verbose", action="store_true")
     parser.add_argument("--perf-profile", action="store_true")
 
