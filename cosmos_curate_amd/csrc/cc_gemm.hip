@@ -97,7 +97,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg) {
+    int c_is_bf16, int nbx, int nwg, int do_remap) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];  // one __shared__ object (G16 4a)
 #define AS(b) (lds + (b) * (BM * BK))
 #define BS(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -111,7 +111,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
   // b ran on XCD b%8; give XCD x the contiguous tile range so its L2
   // sees adjacent tiles.  q = nwg/8, r = nwg%8.
   int orig = blockIdx.x;
-  {
+  if (do_remap) {
     int q = nwg >> 3, r = nwg & 7;
     int xcd = orig & 7, lid = orig >> 3;
     orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
@@ -197,7 +197,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg) {
+    int c_is_bf16, int nbx, int nwg, int do_remap) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];
 #define AS32(b) (lds + (b) * (BM * BK))
 #define BS32(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -207,7 +207,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
   const int waveM = wid >> 1, waveN = wid & 1;
 
   int orig = blockIdx.x;
-  {
+  if (do_remap) {
     int q = nwg >> 3, r = nwg & 7;
     int xcd = orig & 7, lid = orig >> 3;
     orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
@@ -299,27 +299,32 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
   // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
   const bool wide = K < 2048;
+  // XCD-aware remap: helps HBM-bound big-square shapes (+10% at 8k in the
+  // guide's ladder) but was measured NEGATIVE on the L2-resident ViT
+  // shapes (profiles/r01_opt3: prod-with-remap 458 vs no-remap 620 TF on
+  // QKV) — enable only for large outputs that spill L3.
+  const int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
   if (wide) {
     if (act == 1)
       hipLaunchKernelGGL(k_gemm_bf16_w32<1>, grid, block, 0, (hipStream_t)stream,
                          (const __bf16*)A, (const __bf16*)B, C, bias,
                          (const __bf16*)residual, (long)M, (long)N, (long)K,
-                         c_dtype == 1 ? 1 : 0, nbx, nwg);
+                         c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
     else
       hipLaunchKernelGGL(k_gemm_bf16_w32<0>, grid, block, 0, (hipStream_t)stream,
                          (const __bf16*)A, (const __bf16*)B, C, bias,
                          (const __bf16*)residual, (long)M, (long)N, (long)K,
-                         c_dtype == 1 ? 1 : 0, nbx, nwg);
+                         c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
   } else if (act == 1)
     hipLaunchKernelGGL(k_gemm_bf16<1>, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, bias,
                        (const __bf16*)residual, (long)M, (long)N, (long)K,
-                       c_dtype == 1 ? 1 : 0, nbx, nwg);
+                       c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
   else
     hipLaunchKernelGGL(k_gemm_bf16<0>, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, bias,
                        (const __bf16*)residual, (long)M, (long)N, (long)K,
-                       c_dtype == 1 ? 1 : 0, nbx, nwg);
+                       c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
   hipError_t e = hipGetLastError();
   if (timed) cc::timed_end("gemm_bf16", stream, ev0, ev1);
   if (e != hipSuccess)
