@@ -92,7 +92,7 @@ __device__ __forceinline__ bf16x8 frag_read(const __bf16* tile, int row,
   return *(const bf16x8*)(tile + (long)row * BK + slot * 8);
 }
 
-template <int ACT>  // 0 = none, 1 = quick-gelu
+template <int ACT, bool HAS_BIAS, bool HAS_RES>  // ACT 1 = quick-gelu
 __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
@@ -161,23 +161,59 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     buf ^= 1;
   }
 
-  // epilogue: C/D map for 16x16x32: col = lane&15, row = (lane>>4)*4 + reg
+  // epilogue: C/D map for 16x16x32: col = lane&15, row = (lane>>4)*4 + reg.
+  // Interior tiles take the unguarded path: per-element bounds branches
+  // around the residual/bias loads force hipcc into one dependent
+  // load+vmcnt(0) per element (guide §5 trap (c); 458 vs 632 TF measured).
   const long crow_base = bm + waveM * WM + 4 * (lane >> 4);
   const long ccol_base = bn + waveN * WN + (lane & 15);
+  const bool interior = (bm + BM <= M) && (bn + BN <= N);
+  if (interior) {
+#pragma unroll
+    for (int m = 0; m < MFR; m++) {
+#pragma unroll
+      for (int n = 0; n < NFR; n++) {
+        const long col = ccol_base + n * FRAG;
+        float bval = 0.0f;
+        if constexpr (HAS_BIAS) bval = bias[col];
+        // batch the residual loads (independent, one wait) before the
+        // store sequence — an interleaved load/store chain serializes
+        float rv[4];
+        if constexpr (HAS_RES) {
+#pragma unroll
+          for (int r = 0; r < 4; r++)
+            rv[r] = (float)residual[(crow_base + m * FRAG + r) * N + col];
+        }
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const long row = crow_base + m * FRAG + r;
+          float v = acc[m][n][r] + bval;
+          if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+          if constexpr (HAS_RES) v += rv[r];
+          if (c_is_bf16)
+            ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+          else
+            ((float*)C)[row * N + col] = v;
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int m = 0; m < MFR; m++) {
 #pragma unroll
     for (int n = 0; n < NFR; n++) {
       const long col = ccol_base + n * FRAG;
       if (col >= N) continue;
-      const float bval = bias ? bias[col] : 0.0f;
+      float bval = 0.0f;
+      if constexpr (HAS_BIAS) bval = bias[col];
 #pragma unroll
       for (int r = 0; r < 4; r++) {
         const long row = crow_base + m * FRAG + r;
         if (row >= M) continue;
         float v = acc[m][n][r] + bval;
         if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));  // quick-gelu
-        if (residual) v += (float)residual[row * N + col];
+        if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
         else
@@ -192,7 +228,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
 // +19-36% over the 16x16x32 body on the K=768 ViT shapes and +7% on big
 // squares (fewer, denser MFMAs: 2382 vs 2075 TF ceiling); the 16x16 body
 // keeps a small edge at K >= 2048 (patch-embed), so the launcher picks by K.
-template <int ACT>
+template <int ACT, bool HAS_BIAS, bool HAS_RES>
 __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
@@ -251,21 +287,53 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
   }
   // C/D map 32x32x16: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
   const long col0 = bn + waveN * WN + (lane & 31);
+  const long row0 = bm + waveM * WM + 4 * (lane >> 5);
+  const bool interior = (bm + BM <= M) && (bn + BN <= N);
+  if (interior) {
+#pragma unroll
+    for (int m = 0; m < 2; m++) {
+#pragma unroll
+      for (int n = 0; n < 2; n++) {
+        const long col = col0 + n * 32;
+        float bval = 0.0f;
+        if constexpr (HAS_BIAS) bval = bias[col];
+        float rv[16];
+        if constexpr (HAS_RES) {
+#pragma unroll
+          for (int reg = 0; reg < 16; reg++)
+            rv[reg] = (float)residual[
+                (row0 + m * 32 + (reg & 3) + 8 * (reg >> 2)) * N + col];
+        }
+#pragma unroll
+        for (int reg = 0; reg < 16; reg++) {
+          const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
+          float v = acc[m][n][reg] + bval;
+          if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+          if constexpr (HAS_RES) v += rv[reg];
+          if (c_is_bf16)
+            ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+          else
+            ((float*)C)[row * N + col] = v;
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int m = 0; m < 2; m++) {
 #pragma unroll
     for (int n = 0; n < 2; n++) {
       const long col = col0 + n * 32;
       if (col >= N) continue;
-      const float bval = bias ? bias[col] : 0.0f;
+      float bval = 0.0f;
+      if constexpr (HAS_BIAS) bval = bias[col];
 #pragma unroll
       for (int reg = 0; reg < 16; reg++) {
-        const long row =
-            bm + waveM * WM + m * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
         if (row >= M) continue;
         float v = acc[m][n][reg] + bval;
         if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
-        if (residual) v += (float)residual[row * N + col];
+        if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
         else
@@ -299,32 +367,36 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
   // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
   const bool wide = K < 2048;
-  // XCD-aware remap: helps HBM-bound big-square shapes (+10% at 8k in the
-  // guide's ladder) but was measured NEGATIVE on the L2-resident ViT
-  // shapes (profiles/r01_opt3: prod-with-remap 458 vs no-remap 620 TF on
-  // QKV) — enable only for large outputs that spill L3.
+  const bool hb = bias != nullptr;
+  const bool hr = residual != nullptr;
+  // XCD remap only for outputs that spill L3 (measured negative on the
+  // L2-resident ViT shapes, profiles/r01_opt3)
   const int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
-  if (wide) {
-    if (act == 1)
-      hipLaunchKernelGGL(k_gemm_bf16_w32<1>, grid, block, 0, (hipStream_t)stream,
-                         (const __bf16*)A, (const __bf16*)B, C, bias,
-                         (const __bf16*)residual, (long)M, (long)N, (long)K,
-                         c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
-    else
-      hipLaunchKernelGGL(k_gemm_bf16_w32<0>, grid, block, 0, (hipStream_t)stream,
-                         (const __bf16*)A, (const __bf16*)B, C, bias,
-                         (const __bf16*)residual, (long)M, (long)N, (long)K,
-                         c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
-  } else if (act == 1)
-    hipLaunchKernelGGL(k_gemm_bf16<1>, grid, block, 0, (hipStream_t)stream,
-                       (const __bf16*)A, (const __bf16*)B, C, bias,
-                       (const __bf16*)residual, (long)M, (long)N, (long)K,
-                       c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
+#define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
+  hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \
+                     (const __bf16*)A, (const __bf16*)B, C, bias,             \
+                     (const __bf16*)residual, (long)M, (long)N, (long)K,      \
+                     c_dtype == 1 ? 1 : 0, nbx, nwg, remap)
+#define CC_DISPATCH(KER)                                                      \
+  do {                                                                        \
+    if (act == 1) {                                                           \
+      if (hb && hr) CC_LAUNCH_GEMM(KER, 1, true, true);                       \
+      else if (hb) CC_LAUNCH_GEMM(KER, 1, true, false);                       \
+      else if (hr) CC_LAUNCH_GEMM(KER, 1, false, true);                       \
+      else CC_LAUNCH_GEMM(KER, 1, false, false);                              \
+    } else {                                                                  \
+      if (hb && hr) CC_LAUNCH_GEMM(KER, 0, true, true);                       \
+      else if (hb) CC_LAUNCH_GEMM(KER, 0, true, false);                       \
+      else if (hr) CC_LAUNCH_GEMM(KER, 0, false, true);                       \
+      else CC_LAUNCH_GEMM(KER, 0, false, false);                              \
+    }                                                                         \
+  } while (0)
+  if (wide)
+    CC_DISPATCH(k_gemm_bf16_w32);
   else
-    hipLaunchKernelGGL(k_gemm_bf16<0>, grid, block, 0, (hipStream_t)stream,
-                       (const __bf16*)A, (const __bf16*)B, C, bias,
-                       (const __bf16*)residual, (long)M, (long)N, (long)K,
-                       c_dtype == 1 ? 1 : 0, nbx, nwg, remap);
+    CC_DISPATCH(k_gemm_bf16);
+#undef CC_DISPATCH
+#undef CC_LAUNCH_GEMM
   hipError_t e = hipGetLastError();
   if (timed) cc::timed_end("gemm_bf16", stream, ev0, ev1);
   if (e != hipSuccess)

@@ -35,11 +35,15 @@ CLIP_STD = (0.26862954, 0.26130258, 0.27577711)
 class _CLIPImageEmbeddings(torch.nn.Module):
     """Device-side implementation (reference clip.py:36-74 counterpart)."""
 
-    def __init__(self) -> None:
+    def __init__(self, variant: str = "vit_b32") -> None:
         super().__init__()
+        from cosmos_curate_amd.models import clip_weights as cw
+
         hotpath.require_gpu()  # fail loudly before any lazy surprises
         self.device = torch.device("cuda")
-        self.tower = ClipVisionTowerAMD(make_clip_vit_b32_weights()).to(self.device)
+        cfg = cw.CONFIGS[variant]
+        sd = make_clip_vit_b32_weights() if variant == "vit_b32" else cw.make_clip_vit_weights(cfg)
+        self.tower = ClipVisionTowerAMD(sd, cfg).to(self.device)
         self._mean = (np.array(CLIP_MEAN, dtype=np.float32) * 1.0).ctypes
         self._std = (np.array(CLIP_STD, dtype=np.float32) * 1.0).ctypes
         self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
@@ -77,10 +81,15 @@ class _CLIPImageEmbeddings(torch.nn.Module):
 
 
 class CLIPImageEmbeddings(ModelInterface):
-    """ModelInterface wrapper (reference clip.py:77-118)."""
+    """ModelInterface wrapper (reference clip.py:77-118).
 
-    def __init__(self) -> None:
+    variant "vit_b32" (flagship) | "vit_l14" (the reference's own CLIP
+    model id, openai/clip-vit-large-patch14 — BASELINE config #3 class).
+    """
+
+    def __init__(self, variant: str = "vit_b32") -> None:
         super().__init__()
+        self._variant = variant
         self._model: _CLIPImageEmbeddings | None = None
 
     @property
@@ -89,10 +98,12 @@ class CLIPImageEmbeddings(ModelInterface):
 
     @property
     def model_id_names(self) -> list[str]:
+        if self._variant == "vit_l14":
+            return ["openai/clip-vit-large-patch14"]  # reference clip.py:33
         return [_CLIP_MODEL_ID]
 
     def setup(self) -> None:
-        self._model = _CLIPImageEmbeddings()
+        self._model = _CLIPImageEmbeddings(self._variant)
 
     def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
         assert self._model is not None, "setup() not called"

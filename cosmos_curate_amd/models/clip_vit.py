@@ -61,24 +61,44 @@ def _cc_linear(
 
 
 class ClipVisionTowerAMD(torch.nn.Module):
-    """ViT-B/32 vision tower + projection on the MFMA GEMM path."""
+    """CLIP vision tower + projection on the MFMA GEMM path.
 
-    def __init__(self, state_dict: dict[str, torch.Tensor] | None = None) -> None:
+    Parameterized over the tower geometry (clip_weights.VitConfig):
+    ViT-B/32 (flagship, configs #1/#2) and ViT-L/14 (the reference's own
+    CLIP model, models/clip.py:33; BASELINE config #3 class).  The
+    patch-embed GEMM's K = 3*patch^2 is zero-padded to the kernel's
+    64-multiple requirement (588 -> 640 for L/14) — exact, the padded
+    columns multiply zeros.
+    """
+
+    def __init__(
+        self,
+        state_dict: dict[str, torch.Tensor] | None = None,
+        cfg: cw.VitConfig = cw.VIT_B32,
+    ) -> None:
         super().__init__()
-        sd = state_dict if state_dict is not None else cw.make_clip_vit_b32_weights()
+        self.cfg = cfg
+        if state_dict is None:
+            state_dict = (
+                cw.make_clip_vit_b32_weights() if cfg is cw.VIT_B32
+                else cw.make_clip_vit_weights(cfg)
+            )
+        sd = state_dict
         p = "vision_model."
         reg = self.register_buffer
         to_bf = lambda t: t.to(torch.bfloat16).contiguous()  # noqa: E731
 
+        k0 = 3 * cfg.patch * cfg.patch
+        self.patch_k = (k0 + 63) // 64 * 64  # cc_gemm needs K % 64 == 0
         reg("cls_emb", sd[p + "embeddings.class_embedding"].clone())
-        # patch conv as GEMM weight: [768, 3*32*32] with (c,ky,kx) flattening
-        reg("w_patch", to_bf(sd[p + "embeddings.patch_embedding.weight"].reshape(cw.HIDDEN, -1)))
+        w_patch = sd[p + "embeddings.patch_embedding.weight"].reshape(cfg.hidden, k0)
+        if self.patch_k != k0:
+            w_patch = torch.nn.functional.pad(w_patch, (0, self.patch_k - k0))
+        reg("w_patch", to_bf(w_patch))
         reg("pos_emb", sd[p + "embeddings.position_embedding.weight"].clone())
         reg("pre_ln_w", sd[p + "pre_layrnorm.weight"].clone())
         reg("pre_ln_b", sd[p + "pre_layrnorm.bias"].clone())
-        self.layers = len(
-            {k.split(".")[3] for k in sd if k.startswith(p + "encoder.layers.")}
-        )
+        self.layers = cfg.layers
         for i in range(self.layers):
             q = f"{p}encoder.layers.{i}."
             # fused QKV: [3*768, 768] rows ordered (q, k, v)
@@ -99,8 +119,8 @@ class ClipVisionTowerAMD(torch.nn.Module):
         reg("post_ln_w", sd[p + "post_layernorm.weight"].clone())
         reg("post_ln_b", sd[p + "post_layernorm.bias"].clone())
         reg("w_proj", to_bf(sd["visual_projection.weight"]))
-        self.heads = cw.HEADS
-        self.scale = 1.0 / math.sqrt(cw.HIDDEN // cw.HEADS)
+        self.heads = cfg.heads
+        self.scale = 1.0 / math.sqrt(cfg.hidden // cfg.heads)
 
     # the one contraction primitive; tests may monkeypatch this
     def _linear(
@@ -120,29 +140,33 @@ class ClipVisionTowerAMD(torch.nn.Module):
 
     @torch.no_grad()
     def forward(self, pixel_values: torch.Tensor) -> torch.Tensor:
-        """(N,3,224,224) bf16 -> (N,512) f32 L2-normalized embeddings."""
+        """(N,3,224,224) bf16 -> (N,proj) f32 L2-normalized embeddings."""
+        cfg = self.cfg
         x = pixel_values.to(torch.bfloat16)
         n = x.shape[0]
-        g = cw.IMAGE // cw.PATCH  # 7
+        g = cfg.image // cfg.patch
+        k0 = 3 * cfg.patch * cfg.patch
         # patch extraction: (n,c,ph,ky,pw,kx) -> (n,ph,pw,c,ky,kx)
         patches = (
-            x.reshape(n, 3, g, cw.PATCH, g, cw.PATCH)
+            x.reshape(n, 3, g, cfg.patch, g, cfg.patch)
             .permute(0, 2, 4, 1, 3, 5)
-            .reshape(n * g * g, 3 * cw.PATCH * cw.PATCH)
+            .reshape(n * g * g, k0)
         )
-        tok = self._linear(patches, self.w_patch, None).reshape(n, g * g, cw.HIDDEN)
-        cls = self.cls_emb.to(tok.dtype).expand(n, 1, cw.HIDDEN)
-        h = torch.cat([cls, tok], dim=1)  # (n, 50, 768)
+        if self.patch_k != k0:
+            patches = torch.nn.functional.pad(patches, (0, self.patch_k - k0))
+        tok = self._linear(patches, self.w_patch, None).reshape(n, g * g, cfg.hidden)
+        cls = self.cls_emb.to(tok.dtype).expand(n, 1, cfg.hidden)
+        h = torch.cat([cls, tok], dim=1)  # (n, tokens, hidden)
         h = (h.float() + self.pos_emb.unsqueeze(0)).to(torch.bfloat16)
         h = self._ln(h, self.pre_ln_w, self.pre_ln_b)
 
         seq = h.shape[1]
-        hd = cw.HIDDEN // self.heads
+        hd = cfg.hidden // self.heads
         for i in range(self.layers):
             res = h
             y = self._ln(h, getattr(self, f"ln1_w_{i}"), getattr(self, f"ln1_b_{i}"))
             qkv = self._linear(
-                y.reshape(n * seq, cw.HIDDEN),
+                y.reshape(n * seq, cfg.hidden),
                 getattr(self, f"w_qkv_{i}"),
                 getattr(self, f"b_qkv_{i}"),
             ).reshape(n, seq, 3, self.heads, hd)
@@ -152,26 +176,26 @@ class ClipVisionTowerAMD(torch.nn.Module):
             attn = torch.nn.functional.scaled_dot_product_attention(
                 q, k, v, scale=self.scale
             )
-            attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cw.HIDDEN)
+            attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cfg.hidden)
             # residual add fused into the out-proj epilogue
             h = self._linear(
                 attn, getattr(self, f"w_out_{i}"), getattr(self, f"b_out_{i}"),
-                residual=res.reshape(n * seq, cw.HIDDEN),
-            ).reshape(n, seq, cw.HIDDEN)
+                residual=res.reshape(n * seq, cfg.hidden),
+            ).reshape(n, seq, cfg.hidden)
 
             res = h
             y = self._ln(h, getattr(self, f"ln2_w_{i}"), getattr(self, f"ln2_b_{i}"))
             # quick-gelu fused into the fc1 epilogue (transformers CLIP act)
             y = self._linear(
-                y.reshape(n * seq, cw.HIDDEN),
+                y.reshape(n * seq, cfg.hidden),
                 getattr(self, f"w_fc1_{i}"),
                 getattr(self, f"b_fc1_{i}"),
                 act=1,
             )
             h = self._linear(
                 y, getattr(self, f"w_fc2_{i}"), getattr(self, f"b_fc2_{i}"),
-                residual=res.reshape(n * seq, cw.HIDDEN),
-            ).reshape(n, seq, cw.HIDDEN)
+                residual=res.reshape(n * seq, cfg.hidden),
+            ).reshape(n, seq, cfg.hidden)
 
         pooled = self._ln(h[:, 0], self.post_ln_w, self.post_ln_b)
         emb = self._linear(pooled.to(torch.bfloat16), self.w_proj, None).float()

@@ -1,4 +1,4 @@
-"""Deterministic fixed-seed CLIP-ViT-B/32 vision weights.
+"""Deterministic fixed-seed CLIP vision-tower weights (ViT-B/32, ViT-L/14).
 
 HF hub weights are unavailable offline (no network in either container), so
 the rebuild's stand-in for the model-weight download step
@@ -19,15 +19,44 @@ import hashlib
 
 import torch
 
-# ViT-B/32 vision config (transformers CLIPVisionConfig defaults)
-HIDDEN = 768
-LAYERS = 12
-HEADS = 12
-INTERMEDIATE = 3072
-PATCH = 32
-IMAGE = 224
-PROJ = 512
-NUM_POS = (IMAGE // PATCH) ** 2 + 1  # 50
+import dataclasses
+
+
+@dataclasses.dataclass(frozen=True)
+class VitConfig:
+    """Vision-tower geometry (transformers CLIPVisionConfig fields)."""
+
+    name: str
+    hidden: int
+    layers: int
+    heads: int
+    intermediate: int
+    patch: int
+    proj: int
+    image: int = 224
+
+    @property
+    def num_pos(self) -> int:
+        return (self.image // self.patch) ** 2 + 1
+
+
+# ViT-B/32 = transformers CLIPVisionConfig defaults (config #1/#2 embedder)
+VIT_B32 = VitConfig("vit_b32", hidden=768, layers=12, heads=12,
+                    intermediate=3072, patch=32, proj=512)
+# ViT-L/14 = openai/clip-vit-large-patch14 geometry — the model the
+# reference's CLIPImageEmbeddings actually loads (models/clip.py:33) and
+# the L/14-class of BASELINE config #3
+VIT_L14 = VitConfig("vit_l14", hidden=1024, layers=24, heads=16,
+                    intermediate=4096, patch=14, proj=768)
+
+CONFIGS = {"vit_b32": VIT_B32, "vit_l14": VIT_L14}
+
+# module-level aliases for the flagship config (bench flop accounting)
+HIDDEN, LAYERS, HEADS = VIT_B32.hidden, VIT_B32.layers, VIT_B32.heads
+INTERMEDIATE, PATCH, IMAGE, PROJ = (
+    VIT_B32.intermediate, VIT_B32.patch, VIT_B32.image, VIT_B32.proj,
+)
+NUM_POS = VIT_B32.num_pos
 
 
 def _seed_for(name: str) -> int:
@@ -39,32 +68,72 @@ def _randn(name: str, *shape: int, std: float = 0.02) -> torch.Tensor:
     return torch.randn(*shape, generator=g, dtype=torch.float32) * std
 
 
-def make_clip_vit_b32_weights() -> dict[str, torch.Tensor]:
-    """State dict for CLIPVisionModelWithProjection (ViT-B/32)."""
+def make_clip_vit_weights(cfg: VitConfig = VIT_B32) -> dict[str, torch.Tensor]:
+    """State dict for CLIPVisionModelWithProjection of the given geometry."""
     sd: dict[str, torch.Tensor] = {}
     p = "vision_model."
-    sd[p + "embeddings.class_embedding"] = _randn("cls", HIDDEN)
+    tag = cfg.name + ":"
+    sd[p + "embeddings.class_embedding"] = _randn(tag + "cls", cfg.hidden)
     sd[p + "embeddings.patch_embedding.weight"] = _randn(
-        "patch", HIDDEN, 3, PATCH, PATCH
+        tag + "patch", cfg.hidden, 3, cfg.patch, cfg.patch
     )
-    sd[p + "embeddings.position_embedding.weight"] = _randn("pos", NUM_POS, HIDDEN)
+    sd[p + "embeddings.position_embedding.weight"] = _randn(
+        tag + "pos", cfg.num_pos, cfg.hidden
+    )
     # transformers' historical key spelling: pre_layrnorm
-    sd[p + "pre_layrnorm.weight"] = torch.ones(HIDDEN)
-    sd[p + "pre_layrnorm.bias"] = torch.zeros(HIDDEN)
-    for i in range(LAYERS):
+    sd[p + "pre_layrnorm.weight"] = torch.ones(cfg.hidden)
+    sd[p + "pre_layrnorm.bias"] = torch.zeros(cfg.hidden)
+    for i in range(cfg.layers):
         q = f"{p}encoder.layers.{i}."
         for proj in ["q_proj", "k_proj", "v_proj", "out_proj"]:
-            sd[q + f"self_attn.{proj}.weight"] = _randn(f"l{i}.{proj}.w", HIDDEN, HIDDEN)
-            sd[q + f"self_attn.{proj}.bias"] = _randn(f"l{i}.{proj}.b", HIDDEN)
-        sd[q + "layer_norm1.weight"] = torch.ones(HIDDEN)
-        sd[q + "layer_norm1.bias"] = torch.zeros(HIDDEN)
-        sd[q + "layer_norm2.weight"] = torch.ones(HIDDEN)
-        sd[q + "layer_norm2.bias"] = torch.zeros(HIDDEN)
-        sd[q + "mlp.fc1.weight"] = _randn(f"l{i}.fc1.w", INTERMEDIATE, HIDDEN)
-        sd[q + "mlp.fc1.bias"] = _randn(f"l{i}.fc1.b", INTERMEDIATE)
-        sd[q + "mlp.fc2.weight"] = _randn(f"l{i}.fc2.w", HIDDEN, INTERMEDIATE)
-        sd[q + "mlp.fc2.bias"] = _randn(f"l{i}.fc2.b", HIDDEN)
-    sd[p + "post_layernorm.weight"] = torch.ones(HIDDEN)
-    sd[p + "post_layernorm.bias"] = torch.zeros(HIDDEN)
-    sd["visual_projection.weight"] = _randn("vproj", PROJ, HIDDEN)
+            sd[q + f"self_attn.{proj}.weight"] = _randn(
+                f"{tag}l{i}.{proj}.w", cfg.hidden, cfg.hidden
+            )
+            sd[q + f"self_attn.{proj}.bias"] = _randn(f"{tag}l{i}.{proj}.b", cfg.hidden)
+        sd[q + "layer_norm1.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm1.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "layer_norm2.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm2.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "mlp.fc1.weight"] = _randn(f"{tag}l{i}.fc1.w", cfg.intermediate, cfg.hidden)
+        sd[q + "mlp.fc1.bias"] = _randn(f"{tag}l{i}.fc1.b", cfg.intermediate)
+        sd[q + "mlp.fc2.weight"] = _randn(f"{tag}l{i}.fc2.w", cfg.hidden, cfg.intermediate)
+        sd[q + "mlp.fc2.bias"] = _randn(f"{tag}l{i}.fc2.b", cfg.hidden)
+    sd[p + "post_layernorm.weight"] = torch.ones(cfg.hidden)
+    sd[p + "post_layernorm.bias"] = torch.zeros(cfg.hidden)
+    sd["visual_projection.weight"] = _randn(tag + "vproj", cfg.proj, cfg.hidden)
+    return sd
+
+
+def make_clip_vit_b32_weights() -> dict[str, torch.Tensor]:
+    """Flagship (ViT-B/32) weights — original entry point.
+
+    NOTE: generated under the legacy un-tagged seed names so existing
+    golden-pinned tests and the smoke cosine stay bit-identical.
+    """
+    cfg = VIT_B32
+    sd: dict[str, torch.Tensor] = {}
+    p = "vision_model."
+    sd[p + "embeddings.class_embedding"] = _randn("cls", cfg.hidden)
+    sd[p + "embeddings.patch_embedding.weight"] = _randn(
+        "patch", cfg.hidden, 3, cfg.patch, cfg.patch
+    )
+    sd[p + "embeddings.position_embedding.weight"] = _randn("pos", cfg.num_pos, cfg.hidden)
+    sd[p + "pre_layrnorm.weight"] = torch.ones(cfg.hidden)
+    sd[p + "pre_layrnorm.bias"] = torch.zeros(cfg.hidden)
+    for i in range(cfg.layers):
+        q = f"{p}encoder.layers.{i}."
+        for proj in ["q_proj", "k_proj", "v_proj", "out_proj"]:
+            sd[q + f"self_attn.{proj}.weight"] = _randn(f"l{i}.{proj}.w", cfg.hidden, cfg.hidden)
+            sd[q + f"self_attn.{proj}.bias"] = _randn(f"l{i}.{proj}.b", cfg.hidden)
+        sd[q + "layer_norm1.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm1.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "layer_norm2.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm2.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "mlp.fc1.weight"] = _randn(f"l{i}.fc1.w", cfg.intermediate, cfg.hidden)
+        sd[q + "mlp.fc1.bias"] = _randn(f"l{i}.fc1.b", cfg.intermediate)
+        sd[q + "mlp.fc2.weight"] = _randn(f"l{i}.fc2.w", cfg.hidden, cfg.intermediate)
+        sd[q + "mlp.fc2.bias"] = _randn(f"l{i}.fc2.b", cfg.hidden)
+    sd[p + "post_layernorm.weight"] = torch.ones(cfg.hidden)
+    sd[p + "post_layernorm.bias"] = torch.zeros(cfg.hidden)
+    sd["visual_projection.weight"] = _randn("vproj", cfg.proj, cfg.hidden)
     return sd
