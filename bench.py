@@ -33,15 +33,25 @@ import torch
 FRAMES_PER_CLIP = 21  # 10 s @ 30 fps sampled at 2 fps, endpoint included
 SRC_H, SRC_W = 1088, 1920  # H.264 coded size for 1080p
 RES = 224
-# ViT-B/32 geometry for cc_gemm flop accounting (FLOPs = 2*M*N*K)
-_PATCHES, _TOKENS, _HID, _INT, _QKV, _PROJ, _LAYERS = 49, 50, 768, 3072, 2304, 512, 12
 
 
-def gemm_flops_per_frame() -> float:
-    per_layer = 2 * _TOKENS * _HID * _QKV + 2 * _TOKENS * _HID * _HID \
-        + 2 * _TOKENS * _HID * _INT + 2 * _TOKENS * _INT * _HID
+def gemm_flops_per_frame(variant: str = "vit_b32") -> float:
+    """cc_gemm FLOPs (2*M*N*K) per frame for the tower geometry."""
+    from cosmos_curate_amd.models import clip_weights as cw
+
+    cfg = cw.CONFIGS[variant]
+    g = cfg.image // cfg.patch
+    patches, tokens = g * g, g * g + 1
+    patch_k = (3 * cfg.patch * cfg.patch + 63) // 64 * 64  # kernel K padding
+    per_layer = (
+        2 * tokens * cfg.hidden * (3 * cfg.hidden)  # fused qkv
+        + 2 * tokens * cfg.hidden * cfg.hidden      # out proj
+        + 2 * tokens * cfg.hidden * cfg.intermediate
+        + 2 * tokens * cfg.intermediate * cfg.hidden
+    )
     return float(
-        2 * _PATCHES * (3 * 32 * 32) * _HID + _LAYERS * per_layer + 2 * _HID * _PROJ
+        2 * patches * patch_k * cfg.hidden + cfg.layers * per_layer
+        + 2 * cfg.hidden * cfg.proj
     )
 
 
@@ -108,6 +118,11 @@ def main() -> None:
         "the roofline leg still runs eager with HIP-event timing",
     )
     ap.add_argument("--no-graphs", dest="graphs", action="store_false")
+    ap.add_argument(
+        "--model", default="vit_b32", choices=["vit_b32", "vit_l14"],
+        help="embedder tower: vit_b32 (flagship, configs #1/#2) or vit_l14 "
+        "(the reference's CLIP model / config #3 class)",
+    )
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", 1))
@@ -126,7 +141,7 @@ def main() -> None:
     from cosmos_curate_amd.models.clip import _CLIPImageEmbeddings
 
     lib = hotpath.require_gpu()
-    model = _CLIPImageEmbeddings()
+    model = _CLIPImageEmbeddings(args.model)
 
     B = args.clips
     F = B * FRAMES_PER_CLIP
@@ -146,7 +161,7 @@ def main() -> None:
         )
         pixels = model.preprocess_u8(rgb)
         emb = model.tower(pixels)  # (F, 512) f32 unit-norm
-        per_clip = emb.view(B, FRAMES_PER_CLIP, 512).mean(dim=1)
+        per_clip = emb.view(B, FRAMES_PER_CLIP, emb.shape[-1]).mean(dim=1)
         return per_clip / torch.linalg.vector_norm(per_clip, dim=-1, keepdim=True)
 
     def step() -> np.ndarray:
@@ -200,7 +215,7 @@ def main() -> None:
     clips_per_s = total_clips / elapsed
     frames_per_s = clips_per_s * FRAMES_PER_CLIP
 
-    gemm_flops_step = gemm_flops_per_frame() * F
+    gemm_flops_step = gemm_flops_per_frame(args.model) * F
     gemm_flops_total = gemm_flops_step * roofline_steps  # instrumented phase
     achieved = gemm_flops_total / (gemm_ms / 1e3) if gemm_ms > 0 else 0.0
     peak = 2.5e15  # dense bf16 MFMA peak, MI355X_MICROARCH.md (spec; 2495 TF measured)
@@ -226,7 +241,9 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "workload": "split_pipeline 1xMI355X: HIP NV12->RGB/resize + "
-                "CLIP-ViT-B/32 MFMA bf16 (decode seam excluded: no librocdecode in image)",
+                f"CLIP-{'ViT-B/32' if args.model == 'vit_b32' else 'ViT-L/14'} MFMA bf16 "
+                "(decode seam excluded: no librocdecode in image)",
+                "model": args.model,
                 "clips_per_step": B,
                 "frames_per_clip": FRAMES_PER_CLIP,
                 "src": "1080p30 NV12 in HBM",

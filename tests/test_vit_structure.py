@@ -39,6 +39,40 @@ def test_weights_deterministic(weights):
         assert torch.equal(v, again[k]), k
 
 
+def test_tower_l14_matches_transformers_fp32(monkeypatch):
+    """ViT-L/14 geometry (the reference's own CLIP model, clip.py:33):
+    patch-K padding (588->640) + 24-layer tower vs transformers fp32."""
+    pytest.importorskip("transformers")
+    from transformers import CLIPVisionConfig
+    from transformers.models.clip.modeling_clip import CLIPVisionModelWithProjection
+
+    from cosmos_curate_amd.models import clip_weights as cw
+
+    weights = cw.make_clip_vit_weights(cw.VIT_L14)
+    cfg = CLIPVisionConfig(
+        hidden_size=1024, num_hidden_layers=24, num_attention_heads=16,
+        intermediate_size=4096, patch_size=14, projection_dim=768,
+    )
+    ref = CLIPVisionModelWithProjection(cfg)
+    missing, unexpected = ref.load_state_dict(weights, strict=False)
+    assert not [m for m in missing if "position_ids" not in m] and not unexpected
+    ref = ref.float().eval()
+
+    rng = np.random.default_rng(0x14)
+    frames = rng.integers(0, 256, size=(1, 224, 224, 3), dtype=np.uint8)
+    pixels = clip_preprocess(frames)
+    with torch.no_grad():
+        want = ref(pixel_values=torch.from_numpy(pixels)).image_embeds
+        want = (want / torch.linalg.vector_norm(want, dim=-1, keepdim=True)).numpy()
+
+    monkeypatch.setattr(ClipVisionTowerAMD, "_linear", _torch_linear)
+    tower = ClipVisionTowerAMD(weights, cw.VIT_L14)
+    got = tower(torch.from_numpy(pixels)).float().numpy()
+    assert got.shape == (1, 768)
+    cos = float(np.sum(want * got))
+    assert cos >= 0.999, cos
+
+
 def test_tower_matches_transformers_fp32(weights, monkeypatch):
     transformers = pytest.importorskip("transformers")  # noqa: F841
     ref = oracle_vit.build_reference_clip_vision(weights)
