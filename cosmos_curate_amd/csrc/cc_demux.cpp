@@ -11,6 +11,7 @@
 // checks this against oracle/mp4_demux.py on the committed fixtures.
 
 #include <algorithm>
+#include <cstdlib>
 #include <cstring>
 #include <vector>
 
@@ -369,5 +370,265 @@ int cc_demux_packet(cc_demux* d, size_t index, const uint8_t** pkt, size_t* size
 }
 
 void cc_demux_close(cc_demux* d) { delete d; }
+
+}  // extern "C"
+
+// ---- stream-copy clip remux ------------------------------------------
+// Replaces the per-clip ffmpeg re-encode of ClipTranscodingStage
+// (clip_extraction_stages.py:317-441) with a sample-exact remux: the
+// samples whose presentation time falls in [start_s, end_s) are copied
+// bit-for-bit into a fresh minimal MP4 (ftyp/moov/mdat, one chunk),
+// carrying the source avcC, rebased stts/ctts and remapped stss.
+// Requires the span to start on a sync sample (no re-encode fallback —
+// the error is recorded per clip upstream).
+
+namespace {
+
+struct ByteVec {
+  std::vector<uint8_t> v;
+  void u8(uint8_t x) { v.push_back(x); }
+  void u16(uint16_t x) { v.push_back(x >> 8); v.push_back(x & 0xff); }
+  void u32(uint32_t x) {
+    v.push_back(x >> 24); v.push_back((x >> 16) & 0xff);
+    v.push_back((x >> 8) & 0xff); v.push_back(x & 0xff);
+  }
+  void bytes(const uint8_t* p, size_t n) { v.insert(v.end(), p, p + n); }
+  void zeros(size_t n) { v.insert(v.end(), n, 0); }
+};
+
+// box with size patched at close
+struct BoxW {
+  ByteVec& b;
+  size_t at;
+  BoxW(ByteVec& bv, const char* typ) : b(bv), at(bv.v.size()) {
+    b.u32(0);
+    b.bytes((const uint8_t*)typ, 4);
+  }
+  void close() {
+    uint32_t sz = (uint32_t)(b.v.size() - at);
+    b.v[at] = sz >> 24; b.v[at + 1] = (sz >> 16) & 0xff;
+    b.v[at + 2] = (sz >> 8) & 0xff; b.v[at + 3] = sz & 0xff;
+  }
+};
+
+}  // namespace
+
+extern "C" {
+
+void cc_buffer_free(void* p) { free(p); }
+
+int cc_demux_remux_clip(cc_demux* d, double start_s, double end_s,
+                        uint8_t** out, size_t* out_size) {
+  if (!d || !out || !out_size) return cc::set_error(CC_ERR_INVALID, "null arg");
+  if (d->codec != 0 || d->avcc.empty())
+    return cc::set_error(CC_ERR_UNSUPPORTED, "remux supports h264/avcC only");
+  const size_t n = d->dts.size();
+  if (d->offsets.size() != n || d->sizes.size() != n)
+    return cc::set_error(CC_ERR_PARSE, "incomplete sample tables");
+  // select samples by presentation seconds (decode order scan)
+  int64_t lo = -1, hi = -1;  // [lo, hi] inclusive decode-order range
+  for (size_t i = 0; i < n; i++) {
+    double pts = (double)(d->dts[i] + d->cts[i] - d->elst_media_time) / d->timescale;
+    if (pts >= start_s - 1e-9 && pts < end_s - 1e-9) {
+      if (lo < 0) lo = (int64_t)i;
+      hi = (int64_t)i;
+    }
+  }
+  if (lo < 0) return cc::set_error(CC_ERR_INVALID, "empty span");
+  bool lo_sync = d->sync.empty() ||
+                 std::binary_search(d->sync.begin(), d->sync.end(), (uint32_t)(lo + 1));
+  if (!lo_sync)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "span start (sample %lld) is not a sync sample; "
+                         "stream-copy needs a keyframe-aligned span",
+                         (long long)lo);
+  const size_t m = (size_t)(hi - lo + 1);
+
+  // sample data (length-prefixed, copied verbatim) + new tables
+  ByteVec mdat_payload;
+  std::vector<uint32_t> sizes(m);
+  uint64_t total_dur = 0;
+  std::vector<std::pair<uint32_t, uint32_t>> stts;  // (count, delta) runs
+  for (size_t i = 0; i < m; i++) {
+    size_t si = lo + i;
+    sizes[i] = d->sizes[si];
+    mdat_payload.bytes(d->data.data() + d->offsets[si], d->sizes[si]);
+    uint32_t delta;
+    if (si + 1 < n)
+      delta = (uint32_t)(d->dts[si + 1] - d->dts[si]);
+    else if (si > 0)
+      delta = (uint32_t)(d->dts[si] - d->dts[si - 1]);
+    else
+      delta = d->timescale / 30;
+    total_dur += delta;
+    if (!stts.empty() && stts.back().second == delta)
+      stts.back().first++;
+    else
+      stts.push_back({1, delta});
+  }
+
+  auto build_moov = [&](uint32_t chunk_off) {
+    ByteVec b;
+    BoxW moov(b, "moov");
+    {
+      BoxW mvhd(b, "mvhd");
+      b.u32(0);  // version+flags
+      b.u32(0); b.u32(0);
+      b.u32(d->timescale); b.u32((uint32_t)total_dur);
+      b.u32(0x00010000); b.u16(0x0100); b.u16(0); b.u32(0); b.u32(0);
+      const int32_t mat[9] = {0x10000, 0, 0, 0, 0x10000, 0, 0, 0, 0x40000000};
+      for (int32_t x : mat) b.u32((uint32_t)x);
+      b.zeros(24);
+      b.u32(2);  // next track id
+      mvhd.close();
+    }
+    {
+      BoxW trak(b, "trak");
+      {
+        BoxW tkhd(b, "tkhd");
+        b.u32(7);  // version 0, flags enabled
+        b.u32(0); b.u32(0); b.u32(1); b.u32(0); b.u32((uint32_t)total_dur);
+        b.zeros(16);
+        const int32_t mat[9] = {0x10000, 0, 0, 0, 0x10000, 0, 0, 0, 0x40000000};
+        for (int32_t x : mat) b.u32((uint32_t)x);
+        b.u32((uint32_t)d->width << 16); b.u32((uint32_t)d->height << 16);
+        tkhd.close();
+      }
+      {
+        BoxW mdia(b, "mdia");
+        {
+          BoxW mdhd(b, "mdhd");
+          b.u32(0); b.u32(0); b.u32(0);
+          b.u32(d->timescale); b.u32((uint32_t)total_dur);
+          b.u16(0x55c4); b.u16(0);
+          mdhd.close();
+        }
+        {
+          BoxW hdlr(b, "hdlr");
+          b.u32(0); b.u32(0);
+          b.bytes((const uint8_t*)"vide", 4);
+          b.zeros(12);
+          b.bytes((const uint8_t*)"v\0", 2);
+          hdlr.close();
+        }
+        {
+          BoxW minf(b, "minf");
+          {
+            BoxW vmhd(b, "vmhd");
+            b.u32(1); b.zeros(8);
+            vmhd.close();
+          }
+          {
+            BoxW dinf(b, "dinf");
+            BoxW dref(b, "dref");
+            b.u32(0); b.u32(1);
+            BoxW url(b, "url ");
+            b.u32(1);
+            url.close();
+            dref.close();
+            dinf.close();
+          }
+          {
+            BoxW stbl(b, "stbl");
+            {
+              BoxW stsd(b, "stsd");
+              b.u32(0); b.u32(1);
+              BoxW avc1(b, "avc1");
+              b.zeros(6); b.u16(1);
+              b.zeros(16);
+              b.u16((uint16_t)d->width); b.u16((uint16_t)d->height);
+              b.u32(0x00480000); b.u32(0x00480000); b.u32(0);
+              b.u16(1); b.zeros(32);
+              b.u16(0x18); b.u16(0xffff);
+              BoxW avcC(b, "avcC");
+              b.bytes(d->avcc.data(), d->avcc.size());
+              avcC.close();
+              avc1.close();
+              stsd.close();
+            }
+            {
+              BoxW stts_b(b, "stts");
+              b.u32(0); b.u32((uint32_t)stts.size());
+              for (auto& [cnt, delta] : stts) { b.u32(cnt); b.u32(delta); }
+              stts_b.close();
+            }
+            // ctts only if any nonzero offset in range
+            bool any_cts = false;
+            for (size_t i = 0; i < m; i++)
+              if (d->cts[lo + i] != 0) any_cts = true;
+            if (any_cts) {
+              BoxW ctts(b, "ctts");
+              b.u32(0);
+              // run-length encode
+              std::vector<std::pair<uint32_t, int32_t>> runs;
+              for (size_t i = 0; i < m; i++) {
+                int32_t off = d->cts[lo + i];
+                if (!runs.empty() && runs.back().second == off)
+                  runs.back().first++;
+                else
+                  runs.push_back({1, off});
+              }
+              b.u32((uint32_t)runs.size());
+              for (auto& [cnt, off] : runs) { b.u32(cnt); b.u32((uint32_t)off); }
+              ctts.close();
+            }
+            if (!d->sync.empty()) {
+              std::vector<uint32_t> ss;
+              for (uint32_t s : d->sync)
+                if (s >= lo + 1 && s <= (uint32_t)(hi + 1)) ss.push_back(s - (uint32_t)lo);
+              BoxW stss(b, "stss");
+              b.u32(0); b.u32((uint32_t)ss.size());
+              for (uint32_t s : ss) b.u32(s);
+              stss.close();
+            }
+            {
+              BoxW stsz(b, "stsz");
+              b.u32(0); b.u32(0); b.u32((uint32_t)m);
+              for (uint32_t sz : sizes) b.u32(sz);
+              stsz.close();
+            }
+            {
+              BoxW stsc(b, "stsc");
+              b.u32(0); b.u32(1); b.u32(1); b.u32((uint32_t)m); b.u32(1);
+              stsc.close();
+            }
+            {
+              BoxW stco(b, "stco");
+              b.u32(0); b.u32(1); b.u32(chunk_off);
+              stco.close();
+            }
+            stbl.close();
+          }
+          minf.close();
+        }
+        mdia.close();
+      }
+      trak.close();
+    }
+    moov.close();
+    return b.v;
+  };
+
+  static const uint8_t ftyp[] = {0, 0, 0, 28, 'f', 't', 'y', 'p', 'i', 's', 'o',
+                                 'm', 0, 0, 2, 0, 'i', 's', 'o', 'm', 'a', 'v',
+                                 'c', '1', 'm', 'p', '4', '1'};
+  auto moov0 = build_moov(0);
+  uint32_t chunk_off = (uint32_t)(sizeof(ftyp) + moov0.size() + 8);
+  auto moov = build_moov(chunk_off);
+  size_t total = sizeof(ftyp) + moov.size() + 8 + mdat_payload.v.size();
+  uint8_t* buf = (uint8_t*)malloc(total);
+  if (!buf) return cc::set_error(CC_ERR_NOMEM, "remux alloc");
+  size_t pos = 0;
+  memcpy(buf, ftyp, sizeof(ftyp)); pos += sizeof(ftyp);
+  memcpy(buf + pos, moov.data(), moov.size()); pos += moov.size();
+  uint32_t mdat_sz = (uint32_t)(mdat_payload.v.size() + 8);
+  uint8_t hdr[8] = {(uint8_t)(mdat_sz >> 24), (uint8_t)(mdat_sz >> 16),
+                    (uint8_t)(mdat_sz >> 8), (uint8_t)mdat_sz, 'm', 'd', 'a', 't'};
+  memcpy(buf + pos, hdr, 8); pos += 8;
+  memcpy(buf + pos, mdat_payload.v.data(), mdat_payload.v.size());
+  *out = buf;
+  *out_size = total;
+  return CC_OK;
+}
 
 }  // extern "C"

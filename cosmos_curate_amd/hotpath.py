@@ -37,6 +37,10 @@ def _declare(lib: ctypes.CDLL) -> None:
         c.c_void_p, c.c_size_t, c.POINTER(c.c_void_p), c.POINTER(c.c_size_t),
         c.POINTER(c.c_int64), c.POINTER(c.c_int32)]
     lib.cc_demux_close.argtypes = [c.c_void_p]
+    lib.cc_demux_remux_clip.argtypes = [
+        c.c_void_p, c.c_double, c.c_double, c.POINTER(c.c_void_p),
+        c.POINTER(c.c_size_t)]
+    lib.cc_buffer_free.argtypes = [c.c_void_p]
 
     lib.cc_malloc.argtypes = [c.POINTER(c.c_void_p), c.c_size_t]
     lib.cc_free.argtypes = [c.c_void_p]
@@ -164,6 +168,20 @@ class Demuxer:
             )
         )
         return ctypes.string_at(p, sz.value), pts.value, bool(kf.value)
+
+    def remux_clip(self, start_s: float, end_s: float) -> bytes:
+        """Sample-exact stream-copy of the span into a standalone MP4."""
+        buf = ctypes.c_void_p()
+        sz = ctypes.c_size_t()
+        check(
+            self._lib.cc_demux_remux_clip(
+                self._h, start_s, end_s, ctypes.byref(buf), ctypes.byref(sz)
+            )
+        )
+        try:
+            return ctypes.string_at(buf, sz.value)
+        finally:
+            self._lib.cc_buffer_free(buf)
 
     def close(self) -> None:
         if self._h is not None:

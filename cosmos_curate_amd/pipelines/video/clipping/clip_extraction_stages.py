@@ -229,13 +229,16 @@ def _slice_raw_clip(raw: bytes, span: tuple[float, float]) -> bytes:
 
 
 def _remux_mp4_clip(raw: bytes, span: tuple[float, float]) -> bytes:
-    """Sample-exact H.264 mp4 span remux via the in-repo demuxer+writer.
+    """Sample-exact H.264 mp4 span remux (cc_demux_remux_clip).
 
-    Requires the span's first sample to be a sync sample and a B-frame-free
-    stream; otherwise raises (recorded per clip upstream).  Wiring lands
-    with the rocDecode session (the consumer of these clip payloads).
+    Requires the span's first sample to be a sync sample; otherwise the
+    library fails with CC_ERR_UNSUPPORTED (recorded per clip upstream) —
+    no silent re-encode.
     """
-    raise NotImplementedError("mp4 stream-copy remux lands with the rocDecode session")
+    from cosmos_curate_amd import hotpath
+
+    with hotpath.Demuxer(raw) as d:
+        return d.remux_clip(span[0], span[1])
 
 
 class ClipTranscodingStage(CuratorStage):

@@ -74,6 +74,14 @@ int  cc_demux_timestamps(const cc_demux_t* d, float* out, size_t cap, size_t* n)
 int  cc_demux_packet(cc_demux_t* d, size_t index, const uint8_t** pkt,
                      size_t* size, int64_t* pts, int32_t* keyframe);
 void cc_demux_close(cc_demux_t* d);
+/* Sample-exact stream-copy remux of the presentation span [start_s, end_s)
+ * into a standalone MP4 (replaces the per-clip ffmpeg re-encode,
+ * clip_extraction_stages.py:317-441; DESIGN.md "transcode").  Fails with
+ * CC_ERR_UNSUPPORTED when the span's first sample is not a sync sample.
+ * *out is malloc'd; free with cc_buffer_free. */
+int  cc_demux_remux_clip(cc_demux_t* d, double start_s, double end_s,
+                         uint8_t** out, size_t* out_size);
+void cc_buffer_free(void* p);
 
 /* ---- device memory (library-owned buffers; callers may also pass
  * torch-allocated device pointers to the kernels below) ---------------- */

@@ -117,6 +117,37 @@ def test_vit_gpu_vs_oracle_cosine(lib):
     assert np.all(cos >= 0.999), f"embedding cosine vs fp32 oracle: {cos}"
 
 
+def test_vit_l14_gpu_vs_oracle_cosine(lib):
+    """ViT-L/14 (reference's CLIP model geometry) on the MFMA path."""
+    from transformers import CLIPVisionConfig
+    from transformers.models.clip.modeling_clip import CLIPVisionModelWithProjection
+
+    from cosmos_curate_amd.models import clip_weights as cw
+    from cosmos_curate_amd.models.clip_vit import ClipVisionTowerAMD
+    from oracle.color import clip_preprocess
+
+    weights = cw.make_clip_vit_weights(cw.VIT_L14)
+    cfg = CLIPVisionConfig(
+        hidden_size=1024, num_hidden_layers=24, num_attention_heads=16,
+        intermediate_size=4096, patch_size=14, projection_dim=768,
+    )
+    ref = CLIPVisionModelWithProjection(cfg)
+    ref.load_state_dict(weights, strict=False)
+    ref = ref.float().eval()
+
+    rng = np.random.default_rng(0x114)
+    frames = rng.integers(0, 256, size=(2, 224, 224, 3), dtype=np.uint8)
+    pixels = clip_preprocess(frames)
+    with torch.no_grad():
+        want = ref(pixel_values=torch.from_numpy(pixels)).image_embeds
+        want = (want / torch.linalg.vector_norm(want, dim=-1, keepdim=True)).numpy()
+
+    tower = ClipVisionTowerAMD(weights, cw.VIT_L14).cuda()
+    got = tower(torch.from_numpy(pixels).cuda()).cpu().numpy()
+    cos = np.sum(want * got, axis=1)
+    assert np.all(cos >= 0.999), cos
+
+
 def test_clip_model_interface_end_to_end(lib):
     """CLIPImageEmbeddings from u8 frames (models/clip.py:108-118 surface)."""
     from cosmos_curate_amd.models.clip import CLIPImageEmbeddings

@@ -128,8 +128,9 @@ def test_chunk_tasks_fan_out(corpus):
     assert {c.uuid for t in tasks for c in t.video.clips} == {c.uuid for c in v.clips}
 
 
-def test_mp4_transcode_records_error(golden_dir):
-    """H.264 remux path not wired -> per-clip error, never silent."""
+def test_mp4_transcode_stream_copy_and_sync_guard(golden_dir):
+    """H.264 spans stream-copy sample-exact; non-keyframe-aligned spans
+    record a loud per-clip error, never silently re-encode."""
     import pathlib
     import uuid as uuid_mod
 
@@ -139,11 +140,54 @@ def test_mp4_transcode_records_error(golden_dir):
         SplitPipeTask,
         Video,
     )
+    from oracle import mp4_demux
 
     data = (golden_dir / "synth_bframes.mp4").read_bytes()
     v = Video(input_video=pathlib.Path("/synth.mp4"),
               encoded_data=np.frombuffer(data, dtype=np.uint8))
+    # sync samples are 1 and 13 (30 fps): span from 0.0 copies; span from
+    # 0.2 s (sample 7, mid-GOP) must fail loudly
     v.clips.append(Clip(uuid=uuid_mod.uuid4(), source_video="s", span=(0.0, 0.4)))
+    v.clips.append(Clip(uuid=uuid_mod.uuid4(), source_video="s", span=(0.2, 0.4)))
     out = run_pipeline([SplitPipeTask(videos=[v])], [ClipTranscodingStage()],
                        runner=SequentialRunner())
-    assert "transcode" in out[0].video.clips[0].errors
+    good, bad = out[0].video.clips
+    assert not good.errors
+    clip_bytes = bytes(good.encoded_data.resolve())
+    trk = mp4_demux.parse_mp4(clip_bytes)[0]
+    assert len(trk.dts) == 12  # frames 0..11 display in [0.0, 0.4)
+    src = mp4_demux.parse_mp4(data)[0]
+    for j in [0, 5, 11]:
+        so, ss = src.offsets[j], src.sizes[j]
+        co, cs = trk.offsets[j], trk.sizes[j]
+        assert cs == ss and clip_bytes[co:co + cs] == data[so:so + ss]
+    assert "transcode" in bad.errors and "sync" in bad.errors["transcode"]
+
+
+def test_mp4_remux_sintel_head_span(golden_dir):
+    """Reference fixture (single-IDR): [0,10 s) span copies sample-exact."""
+    import pathlib
+
+    from cosmos_curate_amd import hotpath
+    from oracle import mp4_demux
+
+    src_path = pathlib.Path(
+        "/root/reference/tests/cosmos_curate/pipelines/video/data/test_video_30s.mp4"
+    )
+    if not src_path.exists():
+        pytest.skip("reference fixtures not present (GPU box)")
+    data = src_path.read_bytes()
+    src = mp4_demux.parse_mp4(data)[0]
+    with hotpath.Demuxer(data) as d:
+        clip = d.remux_clip(0.0, 10.0)
+        with pytest.raises(RuntimeError, match="sync"):
+            d.remux_clip(10.0, 20.0)  # only sample 1 is IDR in this encode
+    trk = mp4_demux.parse_mp4(clip)[0]
+    assert len(trk.dts) == 240
+    pts = trk.pts_seconds_sorted()
+    assert abs(pts[0]) < 1e-6 and abs(pts[-1] - 239 / 24) < 1e-4
+    for j in [0, 100, 239]:
+        so, ss = src.offsets[j], src.sizes[j]
+        co, cs = trk.offsets[j], trk.sizes[j]
+        assert cs == ss and clip[co:co + cs] == data[so:so + ss]
+    assert mp4_demux.annexb_packets(clip, trk)[0][:4] == b"\x00\x00\x00\x01"
