@@ -273,6 +273,89 @@ __global__ __launch_bounds__(256, 2) void k_gemm_v4(
 #undef BS4
 }
 
+
+// ---- V5/V6: 8-wave big tiles (256x128 / 128x256), 2-buffer ----
+// Halves LDS-staged bytes per flop vs 128x128 (the L2-bandwidth bound of
+// the staging loop) at unchanged waves/SIMD (512 threads, 1 WG/CU).
+template <int TBM, int TBN, int WGM, int WGN>
+__global__ __launch_bounds__(512, 1) void k_gemm_big(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16) {
+  __shared__ __bf16 lds[(TBM + TBN) * BK * 2];
+  __bf16* const As0 = lds;
+  __bf16* const Bs0 = lds + 2 * TBM * BK;
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;  // 8 waves
+  const int waveM = wid / WGN, waveN = wid % WGN;                // WGM x WGN
+  const long bm = (long)blockIdx.y * TBM, bn = (long)blockIdx.x * TBN;
+  // each of 8 waves stages TBM/8 rows of A and TBN/8 rows of B per tile
+  const long arow0 = bm + (TBM / 8) * wid;
+  const long brow0 = bn + (TBN / 8) * wid;
+
+  f32x16 acc[2][2] = {};  // per-wave 64x64 as 2x2 of 32x32
+  const long KT = K / BK;
+  auto stage_rows = [&](const __bf16* src, long ld, long row0, long limit,
+                        long k0, __bf16* dst, int nrows) {
+    const int lrow8 = lane >> 3, slot = lane & 7, gk16 = slot ^ lrow8;
+    for (int j = 0; j < nrows / 8; j++) {
+      long grow = row0 + j * 8 + lrow8;
+      grow = grow < 0 ? 0 : (grow >= limit ? limit - 1 : grow);
+      const __bf16* gptr = src + grow * ld + k0 + (long)gk16 * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gptr,
+          (__attribute__((address_space(3))) unsigned int*)(dst + j * 8 * BK),
+          16, 0, 0);
+    }
+  };
+  stage_rows(A, K, arow0, M, 0, As0 + (TBM / 8) * wid * BK, TBM / 8);
+  stage_rows(B, K, brow0, N, 0, Bs0 + (TBN / 8) * wid * BK, TBN / 8);
+  int buf = 0;
+  for (long kt = 0; kt < KT; ++kt) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (kt + 1 < KT) {
+      const long k0 = (kt + 1) * BK;
+      stage_rows(A, K, arow0, M, k0, As0 + (buf ^ 1) * TBM * BK + (TBM / 8) * wid * BK, TBM / 8);
+      stage_rows(B, K, brow0, N, k0, Bs0 + (buf ^ 1) * TBN * BK + (TBN / 8) * wid * BK, TBN / 8);
+    }
+    const __bf16* At = As0 + buf * TBM * BK;
+    const __bf16* Bt = Bs0 + buf * TBN * BK;
+    const int arow = waveM * 64 + (lane & 31);
+    const int brow = waveN * 64 + (lane & 31);
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      const int k16 = (kk >> 3) + (lane >> 5);
+      bf16x8 a0 = frag_read(At, arow, k16);
+      bf16x8 a1 = frag_read(At, arow + 32, k16);
+      bf16x8 b0 = frag_read(Bt, brow, k16);
+      bf16x8 b1 = frag_read(Bt, brow + 32, k16);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    buf ^= 1;
+  }
+  const long col0 = bn + waveN * 64 + (lane & 31);
+  const long row0 = bm + waveM * 64 + 4 * (lane >> 5);
+#pragma unroll
+  for (int m = 0; m < 2; m++)
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const long col = col0 + n * 32;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
+        if (row >= M) continue;
+        float v = acc[m][n][reg];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+}
+
 }  // namespace
 
 extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* C,
@@ -293,8 +376,18 @@ extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* 
     hipLaunchKernelGGL(k_gemm_v4, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, (long)M, (long)N,
                        (long)K, c_dtype == 1);
-  else
-    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4");
+  else if (variant == 5) {
+    dim3 g5((N + 127) / 128, (M + 255) / 256);
+    hipLaunchKernelGGL((k_gemm_big<256, 128, 4, 2>), g5, dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A, (const __bf16*)B,
+                       C, (long)M, (long)N, (long)K, c_dtype == 1);
+  } else if (variant == 6) {
+    dim3 g6((N + 255) / 256, (M + 127) / 128);
+    hipLaunchKernelGGL((k_gemm_big<128, 256, 2, 4>), g6, dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A, (const __bf16*)B,
+                       C, (long)M, (long)N, (long)K, c_dtype == 1);
+  } else
+    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4|5|6");
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return cc::set_error(CC_ERR_HIP, "%s", hipGetErrorString(e));
   return CC_OK;
