@@ -31,6 +31,7 @@ SOURCES: list[tuple[str, list[str]]] = [
     ("cc_pixel.hip", [ARCH, "-ffp-contract=off"]),
     ("cc_gemm.hip", [ARCH]),
     ("cc_dedup.hip", [ARCH]),
+    ("cc_ln.hip", [ARCH]),
 ]
 
 

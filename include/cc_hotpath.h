@@ -156,6 +156,11 @@ int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
                     const float* bias, int c_dtype, int act,
                     const void* residual, uint64_t stream);
 
+/* ---- fused bf16 LayerNorm (replaces torch layer_norm in the ViT
+ * forward; f32 stats/affine, H multiple of 128). */
+int cc_layernorm_bf16(const void* x, const void* w, const void* b, void* y,
+                      int64_t M, int64_t H, float eps, uint64_t stream);
+
 /* ---- semantic dedup -------------------------------------------------
  * Strict-upper-triangular max-cosine scan (SemDedupActor.dedup,
  * pipelines/video/dedup/dedup_actor.py:315-460): for each row j of the

@@ -117,6 +117,34 @@ def test_vit_gpu_vs_oracle_cosine(lib):
     assert np.all(cos >= 0.999), f"embedding cosine vs fp32 oracle: {cos}"
 
 
+@pytest.mark.parametrize("M,H", [(1, 128), (333, 768), (1024, 1024), (50, 4096)])
+def test_layernorm_kernel_vs_torch(lib, M, H):
+    torch.manual_seed(M + H)
+    x = (torch.randn(M, H) * 2).to(torch.bfloat16).cuda()
+    w = torch.randn(H).float().cuda()
+    b = torch.randn(H).float().cuda()
+    out = torch.empty_like(x)
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(
+        lib.cc_layernorm_bf16(x.data_ptr(), w.data_ptr(), b.data_ptr(),
+                              out.data_ptr(), M, H, ctypes_float(1e-5), stream)
+    )
+    torch.cuda.synchronize()
+    want = torch.nn.functional.layer_norm(x.float(), (H,), w, b, eps=1e-5).to(
+        torch.bfloat16
+    )
+    torch.testing.assert_close(out, want, rtol=2e-2, atol=2e-2)
+    # element-exact in most positions (same f32 math up to reduction order)
+    frac_diff = (out != want).float().mean().item()
+    assert frac_diff < 0.02, frac_diff
+
+
+def ctypes_float(v):
+    import ctypes
+
+    return ctypes.c_float(v)
+
+
 def test_vit_l14_gpu_vs_oracle_cosine(lib):
     """ViT-L/14 (reference's CLIP model geometry) on the MFMA path."""
     from transformers import CLIPVisionConfig
