@@ -32,6 +32,7 @@ SOURCES: list[tuple[str, list[str]]] = [
     ("cc_gemm.hip", [ARCH]),
     ("cc_dedup.hip", [ARCH]),
     ("cc_ln.hip", [ARCH]),
+    ("cc_attn.hip", [ARCH]),
 ]
 
 

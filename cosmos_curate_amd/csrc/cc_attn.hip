@@ -1,0 +1,193 @@
+// Fused short-sequence multi-head attention for the ViT encoder.
+//
+// Replaces torch-rocm's sdpa (attn_fwd, ~94 TF/s at the bench shape) plus
+// the qkv-slice permute copies around it: reads Q/K/V straight out of the
+// fused-QKV GEMM output (layout [n*seq, 3*H], rows tokens), computes
+// O = softmax(Q K^T / sqrt(hd)) V per (frame, head) entirely in one
+// workgroup (LDS-resident, seq <= 64, hd = 64), and writes O contiguous
+// [n*seq, H] — the exact input layout of the out-projection GEMM.
+// ViT-B/32: seq=50, heads=12, n=frames -> n*12 workgroups.
+// (ViT-L/14's seq=257 exceeds the LDS tile; it stays on torch sdpa.)
+//
+// Numerics: f32 accumulation and softmax (matches sdpa's f32 softmax on
+// bf16 inputs); probabilities round to bf16 before PV (flash-style),
+// covered by the end-to-end cosine tests.
+
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#include "cc_common.hpp"
+#include "cc_timing.hpp"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int SMAX = 64;   // padded sequence tile
+constexpr int HD = 64;     // head dim
+constexpr int LD = 72;     // LDS row stride (pad 8 elements vs 64 banks)
+
+__global__ __launch_bounds__(256, 2) void k_attn_small(
+    const __bf16* __restrict__ qkv, __bf16* __restrict__ out, long n_frames,
+    int seq, int heads, int hidden, float scale) {
+  // one workgroup per (frame, head)
+  const long fh = blockIdx.x;
+  const long frame = fh / heads;
+  const int head = fh % heads;
+  if (frame >= n_frames) return;
+
+  __shared__ __bf16 lds[4 * SMAX * LD];  // Q, K, Vt, P
+  __bf16* Q = lds;
+  __bf16* K = lds + SMAX * LD;
+  __bf16* Vt = lds + 2 * SMAX * LD;
+  __bf16* P = lds + 3 * SMAX * LD;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;  // 4 waves; wave w owns S rows [16w, 16w+16)
+
+  // ---- load Q/K/V tiles: thread t handles row t>>2, cols (t&3)*16.. ----
+  {
+    const int row = tid >> 2;
+    const int d0 = (tid & 3) * 16;
+    const long base = ((frame * seq + row) * 3) * (long)hidden + (long)head * HD;
+    bf16x8 z = {};
+    bf16x8 q0 = z, q1 = z, k0 = z, k1 = z, v0 = z, v1 = z;
+    if (row < seq) {
+      q0 = *(const bf16x8*)(qkv + base + d0);
+      q1 = *(const bf16x8*)(qkv + base + d0 + 8);
+      k0 = *(const bf16x8*)(qkv + base + hidden + d0);
+      k1 = *(const bf16x8*)(qkv + base + hidden + d0 + 8);
+      v0 = *(const bf16x8*)(qkv + base + 2 * hidden + d0);
+      v1 = *(const bf16x8*)(qkv + base + 2 * hidden + d0 + 8);
+    }
+    *(bf16x8*)(Q + row * LD + d0) = q0;
+    *(bf16x8*)(Q + row * LD + d0 + 8) = q1;
+    *(bf16x8*)(K + row * LD + d0) = k0;
+    *(bf16x8*)(K + row * LD + d0 + 8) = k1;
+    // V transposed: Vt[d][row] = V[row][d]
+#pragma unroll
+    for (int e = 0; e < 8; e++) {
+      Vt[(d0 + e) * LD + row] = v0[e];
+      Vt[(d0 + 8 + e) * LD + row] = v1[e];
+    }
+  }
+  __syncthreads();
+
+  // ---- S = Q K^T * scale, rows [16w,16w+16), cols 0..63 ----
+  f32x4 acc[4] = {};
+  {
+    const int qrow = 16 * wid + (lane & 15);
+    const int k0e = 8 * (lane >> 4);
+#pragma unroll
+    for (int kk = 0; kk < HD; kk += 32) {
+      bf16x8 qf = *(const bf16x8*)(Q + qrow * LD + kk + k0e);
+#pragma unroll
+      for (int n = 0; n < 4; n++) {
+        bf16x8 kf = *(const bf16x8*)(K + (n * 16 + (lane & 15)) * LD + kk + k0e);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc[n], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- masked, scaled softmax over cols (per row) ----
+  // acc value (row = 16w + (lane>>4)*4 + reg, col = n*16 + lane&15)
+  float pr[4][4];  // probabilities, same layout
+  {
+    const int colr = lane & 15;
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      float mx = -1e30f;
+      float sv[4];
+#pragma unroll
+      for (int nn = 0; nn < 4; nn++) {
+        float s = acc[nn][reg] * scale;
+        if (nn * 16 + colr >= seq) s = -1e30f;
+        sv[nn] = s;
+        mx = fmaxf(mx, s);
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off));
+      float sum = 0.0f;
+#pragma unroll
+      for (int nn = 0; nn < 4; nn++) {
+        float e = __expf(sv[nn] - mx);
+        if (nn * 16 + colr >= seq) e = 0.0f;
+        sv[nn] = e;
+        sum += e;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
+      const float inv = 1.0f / sum;
+#pragma unroll
+      for (int nn = 0; nn < 4; nn++) pr[nn][reg] = sv[nn] * inv;
+    }
+  }
+  // write P (bf16) to LDS for the PV matmul
+  {
+    const int colr = lane & 15;
+    const int rbase = 16 * wid + 4 * (lane >> 4);
+#pragma unroll
+    for (int nn = 0; nn < 4; nn++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++)
+        P[(rbase + reg) * LD + nn * 16 + colr] = (__bf16)pr[nn][reg];
+  }
+  __syncthreads();
+
+  // ---- O = P Vt^T : rows [16w,16w+16) of O, cols d = 0..63 ----
+  f32x4 oacc[4] = {};
+  {
+    const int prow = 16 * wid + (lane & 15);
+    const int j0 = 8 * (lane >> 4);
+#pragma unroll
+    for (int kk = 0; kk < SMAX; kk += 32) {
+      bf16x8 pf = *(const bf16x8*)(P + prow * LD + kk + j0);
+#pragma unroll
+      for (int n = 0; n < 4; n++) {
+        bf16x8 vf = *(const bf16x8*)(Vt + (n * 16 + (lane & 15)) * LD + kk + j0);
+        oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, oacc[n], 0, 0, 0);
+      }
+    }
+  }
+  // ---- store O: (row = 16w + (lane>>4)*4 + reg, d = n*16 + lane&15) ----
+  {
+    const int dcol = lane & 15;
+    const int rbase = 16 * wid + 4 * (lane >> 4);
+#pragma unroll
+    for (int nn = 0; nn < 4; nn++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int row = rbase + reg;
+        if (row < seq)
+          out[(frame * seq + row) * (long)hidden + head * HD + nn * 16 + dcol] =
+              (__bf16)oacc[nn][reg];
+      }
+  }
+}
+
+}  // namespace
+
+extern "C" int cc_attn_small(const void* qkv, void* out, int64_t n_frames,
+                             int seq, int heads, int hidden, float scale,
+                             uint64_t stream) {
+  if (!qkv || !out || n_frames <= 0 || seq <= 0)
+    return cc::set_error(CC_ERR_INVALID, "bad attn args");
+  if (seq > SMAX || hidden != heads * HD)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "cc_attn_small needs seq <= 64 and hd == 64");
+  dim3 block(256), grid((unsigned)(n_frames * heads));
+  hipEvent_t ev0, ev1;
+  bool timed = cc::timed_begin(stream, &ev0, &ev1);
+  hipLaunchKernelGGL(k_attn_small, grid, block, 0, (hipStream_t)stream,
+                     (const __bf16*)qkv, (__bf16*)out, (long)n_frames, seq,
+                     heads, hidden, scale);
+  hipError_t e = hipGetLastError();
+  if (timed) cc::timed_end("attn_small", stream, ev0, ev1);
+  if (e != hipSuccess)
+    return cc::set_error(CC_ERR_HIP, "attn launch: %s", hipGetErrorString(e));
+  return CC_OK;
+}

@@ -71,6 +71,9 @@ def _declare(lib: ctypes.CDLL) -> None:
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64, c.c_int64,
         c.c_void_p, c.c_int, c.c_int, c.c_void_p, c.c_uint64]
 
+    lib.cc_attn_small.argtypes = [
+        c.c_void_p, c.c_void_p, c.c_int64, c.c_int, c.c_int, c.c_int,
+        c.c_float, c.c_uint64]
     lib.cc_layernorm_bf16.argtypes = [
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64,
         c.c_float, c.c_uint64]

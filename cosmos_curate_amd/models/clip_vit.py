@@ -181,18 +181,36 @@ class ClipVisionTowerAMD(torch.nn.Module):
         for i in range(self.layers):
             res = h
             y = self._ln(h, getattr(self, f"ln1_w_{i}"), getattr(self, f"ln1_b_{i}"))
-            qkv = self._linear(
+            qkv_flat = self._linear(
                 y.reshape(n * seq, cfg.hidden),
                 getattr(self, f"w_qkv_{i}"),
                 getattr(self, f"b_qkv_{i}"),
-            ).reshape(n, seq, 3, self.heads, hd)
-            q = qkv[:, :, 0].permute(0, 2, 1, 3)  # (n, heads, seq, hd)
-            k = qkv[:, :, 1].permute(0, 2, 1, 3)
-            v = qkv[:, :, 2].permute(0, 2, 1, 3)
-            attn = torch.nn.functional.scaled_dot_product_attention(
-                q, k, v, scale=self.scale
             )
-            attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cfg.hidden)
+            if qkv_flat.is_cuda and seq <= 64 and hd == 64:
+                # fused LDS-resident attention straight off the QKV GEMM
+                # output (csrc/cc_attn.hip) — no permute copies
+                lib = hotpath.require_gpu()
+                attn = torch.empty(
+                    (n * seq, cfg.hidden), dtype=torch.bfloat16,
+                    device=qkv_flat.device,
+                )
+                stream = torch.cuda.current_stream(qkv_flat.device).cuda_stream
+                hotpath.check(
+                    lib.cc_attn_small(
+                        qkv_flat.data_ptr(), attn.data_ptr(), n, seq,
+                        self.heads, cfg.hidden,
+                        __import__("ctypes").c_float(self.scale), stream,
+                    )
+                )
+            else:
+                qkv = qkv_flat.reshape(n, seq, 3, self.heads, hd)
+                q = qkv[:, :, 0].permute(0, 2, 1, 3)  # (n, heads, seq, hd)
+                k = qkv[:, :, 1].permute(0, 2, 1, 3)
+                v = qkv[:, :, 2].permute(0, 2, 1, 3)
+                attn = torch.nn.functional.scaled_dot_product_attention(
+                    q, k, v, scale=self.scale
+                )
+                attn = attn.permute(0, 2, 1, 3).reshape(n * seq, cfg.hidden)
             # residual add fused into the out-proj epilogue
             h = self._linear(
                 attn, getattr(self, f"w_out_{i}"), getattr(self, f"b_out_{i}"),

@@ -156,6 +156,13 @@ int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
                     const float* bias, int c_dtype, int act,
                     const void* residual, uint64_t stream);
 
+/* ---- fused short-sequence attention (ViT encoder; replaces torch sdpa
+ * + the qkv permute copies).  qkv = the fused-QKV GEMM output
+ * [n*seq, 3*hidden] bf16; out [n*seq, hidden] bf16; seq <= 64, head dim
+ * 64. */
+int cc_attn_small(const void* qkv, void* out, int64_t n_frames, int seq,
+                  int heads, int hidden, float scale, uint64_t stream);
+
 /* ---- fused bf16 LayerNorm (replaces torch layer_norm in the ViT
  * forward; f32 stats/affine, H multiple of 128). */
 int cc_layernorm_bf16(const void* x, const void* w, const void* b, void* y,

@@ -117,6 +117,33 @@ def test_vit_gpu_vs_oracle_cosine(lib):
     assert np.all(cos >= 0.999), f"embedding cosine vs fp32 oracle: {cos}"
 
 
+@pytest.mark.parametrize("n,seq,heads", [(3, 50, 12), (1, 1, 2), (5, 64, 4), (2, 33, 8)])
+def test_attn_small_vs_torch_sdpa(lib, n, seq, heads):
+    import ctypes
+
+    import math
+
+    hd = 64
+    hidden = heads * hd
+    torch.manual_seed(n * 100 + seq)
+    qkv = torch.randn(n * seq, 3 * hidden).to(torch.bfloat16).cuda()
+    out = torch.empty((n * seq, hidden), dtype=torch.bfloat16, device="cuda")
+    scale = 1.0 / math.sqrt(hd)
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(
+        lib.cc_attn_small(qkv.data_ptr(), out.data_ptr(), n, seq, heads,
+                          hidden, ctypes.c_float(scale), stream)
+    )
+    torch.cuda.synchronize()
+    q, k, v = (
+        qkv.reshape(n, seq, 3, heads, hd)[:, :, i].permute(0, 2, 1, 3).float()
+        for i in range(3)
+    )
+    want = torch.nn.functional.scaled_dot_product_attention(q, k, v, scale=scale)
+    want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
+    torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
+
+
 @pytest.mark.parametrize("M,H", [(1, 128), (333, 768), (1024, 1024), (50, 4096)])
 def test_layernorm_kernel_vs_torch(lib, M, H):
     torch.manual_seed(M + H)
