@@ -358,6 +358,112 @@ __global__ __launch_bounds__(512, 1) void k_gemm_big(
 
 }  // namespace
 
+// ---- V7: half-K ring pipeline ----
+// 128x128 tile, 4-slot LDS ring of 32-deep K chunks (16 KB/slot, 64 KB
+// total -> still 2 WG/CU), counted vmcnt + raw barrier per chunk: the glds
+// queue never drains (prefetch distance 3 chunks).  Trades the 2-buffer
+// vmcnt(0) drain for more barriers + a narrower (64 B-row) LDS image.
+namespace v7 {
+constexpr int CB = 32;  // chunk K depth
+
+__device__ __forceinline__ void stage_chunk32(const __bf16* __restrict__ src,
+                                              long ld, long row0, long limit,
+                                              long k0, __bf16* dst, int lane) {
+  // 32 rows x 32 k (64 B rows): lane l -> row l>>2, 16B slot (l&3)^((l>>2)&3)
+  const int r = lane >> 2;
+  const int gk16 = (lane & 3) ^ (r & 3);
+#pragma unroll
+  for (int j = 0; j < 2; j++) {  // 2 x 1KB = 32 rows
+    long grow = row0 + j * 16 + r;
+    grow = grow < 0 ? 0 : (grow >= limit ? limit - 1 : grow);
+    const __bf16* g = src + grow * ld + k0 + (long)gk16 * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(dst + j * 16 * CB),
+        16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ bf16x8 frag32(const __bf16* t, int row, int k16) {
+  return *(const bf16x8*)(t + (long)row * CB + ((k16 ^ (row & 3)) * 8));
+}
+
+__global__ __launch_bounds__(256, 2) void k_gemm_v7(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16) {
+  __shared__ __bf16 lds[4 * (128 + 128) * CB];  // ring[4]: A 128x32 + B 128x32
+#define A7(s) (lds + (s) * (128 * CB))
+#define B7(s) (lds + 4 * (128 * CB) + (s) * (128 * CB))
+  const int tid = threadIdx.x, lane = tid & 63, wid = tid >> 6;
+  const int waveM = wid >> 1, waveN = wid & 1;
+  const long bm = (long)blockIdx.y * 128, bn = (long)blockIdx.x * 128;
+  const long arow0 = bm + 32 * wid, brow0 = bn + 32 * wid;
+
+  f32x16 acc[2][2] = {};
+  const long NC = K / CB;
+  // prologue: stage chunks 0..2 (ring slots 0..2)
+  for (int c = 0; c < 3 && c < NC; c++) {
+    stage_chunk32(A, K, arow0, M, (long)c * CB, A7(c) + 32 * wid * CB, lane);
+    stage_chunk32(B, K, brow0, N, (long)c * CB, B7(c) + 32 * wid * CB, lane);
+  }
+  for (long c = 0; c < NC; ++c) {
+    const int slot = c & 3;
+    // wait chunk c landed; leave up to 2 younger stages (8 glds) in flight
+    if (c + 1 < NC)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (c + 3 < NC) {  // stage chunk c+3 into slot (c+3)&3 = (c-1)&3 (free)
+      const long k0 = (c + 3) * CB;
+      const int ns = (c + 3) & 3;
+      stage_chunk32(A, K, arow0, M, k0, A7(ns) + 32 * wid * CB, lane);
+      stage_chunk32(B, K, brow0, N, k0, B7(ns) + 32 * wid * CB, lane);
+    }
+    const __bf16* At = A7(slot);
+    const __bf16* Bt = B7(slot);
+    const int ar = waveM * 64 + (lane & 31);
+    const int br = waveN * 64 + (lane & 31);
+#pragma unroll
+    for (int kk = 0; kk < CB; kk += 16) {
+      const int k16 = (kk >> 3) + (lane >> 5);
+      bf16x8 a0 = frag32(At, ar, k16);
+      bf16x8 a1 = frag32(At, ar + 32, k16);
+      bf16x8 b0 = frag32(Bt, br, k16);
+      bf16x8 b1 = frag32(Bt, br + 32, k16);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    // no tail barrier: slot (c+3)&3's last readers were chunk c-1, all past
+    // this chunk's top barrier before the stage above was issued.
+  }
+  const long col0 = bn + waveN * 64 + (lane & 31);
+  const long row0 = bm + waveM * 64 + 4 * (lane >> 5);
+#pragma unroll
+  for (int m = 0; m < 2; m++)
+#pragma unroll
+    for (int n = 0; n < 2; n++) {
+      const long col = col0 + n * 32;
+      if (col >= N) continue;
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
+        if (row >= M) continue;
+        float v = acc[m][n][reg];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+#undef A7
+#undef B7
+}
+}  // namespace v7
+
 extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* C,
                                int64_t M, int64_t N, int64_t K, int c_dtype,
                                uint64_t stream) {
@@ -386,8 +492,13 @@ extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* 
     hipLaunchKernelGGL((k_gemm_big<128, 256, 2, 4>), g6, dim3(512), 0,
                        (hipStream_t)stream, (const __bf16*)A, (const __bf16*)B,
                        C, (long)M, (long)N, (long)K, c_dtype == 1);
+  } else if (variant == 7) {
+    if (K % 32 != 0) return cc::set_error(CC_ERR_UNSUPPORTED, "K%%32");
+    hipLaunchKernelGGL(v7::k_gemm_v7, grid, block, 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C, (long)M,
+                       (long)N, (long)K, c_dtype == 1);
   } else
-    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4|5|6");
+    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4|5|6|7");
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return cc::set_error(CC_ERR_HIP, "%s", hipGetErrorString(e));
   return CC_OK;
