@@ -70,7 +70,7 @@ def make_nv12_batch(n_frames: int, seed: int) -> tuple[np.ndarray, np.ndarray]:
     return y, uv
 
 
-def run_cpu_baseline(n_clips: int = 2) -> dict:
+def run_cpu_baseline(n_clips: int = 8) -> dict:
     """Oracle (CPU restatement) on the same per-clip work, host cores."""
     from oracle import color as ocolor
     from oracle import vit as oracle_vit
