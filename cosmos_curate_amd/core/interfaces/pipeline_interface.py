@@ -39,10 +39,19 @@ class PipelineExecutionError(RuntimeError):
 
 def _build_pipeline_stage_specs(
     stages: Sequence[CuratorStage | CuratorStageSpec],
+    stage_save_config=None,
 ) -> list[CuratorStageSpec]:
-    """Normalize to specs with defaults (pipeline_interface.py:255-275)."""
+    """Normalize to specs with defaults (pipeline_interface.py:255-275);
+    optionally wrap stages for --stage-save (:259, stage_replay harness)."""
+    from cosmos_curate_amd.core.utils.stage_replay import wrap_stage_for_save
+
     specs: list[CuratorStageSpec] = []
     for stage in stages:
+        stage = (
+            wrap_stage_for_save(stage, stage_save_config)
+            if isinstance(stage, CuratorStage)
+            else stage
+        )
         if isinstance(stage, CuratorStage):
             specs.append(CuratorStageSpec(stage))
         elif isinstance(stage, CuratorStageSpec):
@@ -63,12 +72,13 @@ def run_pipeline(
     stages: Sequence[CuratorStage | CuratorStageSpec],
     model_weights_prefix: str = MODEL_WEIGHTS_PREFIX,
     runner: RunnerInterface | None = None,
+    stage_save_config=None,
     args: argparse.Namespace | None = None,
 ) -> list[T]:
     """Run the pipeline (pipeline_interface.py:281-329 contract)."""
     if runner is None:
         runner = SequentialRunner()
-    stage_specs = _build_pipeline_stage_specs(stages)
+    stage_specs = _build_pipeline_stage_specs(stages, stage_save_config)
     execution_mode = getattr(args, "execution_mode", "AUTO") if args is not None else "AUTO"
     try:
         output_tasks = runner.run(input_tasks, stage_specs, model_weights_prefix, execution_mode)

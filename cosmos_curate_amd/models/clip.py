@@ -44,8 +44,6 @@ class _CLIPImageEmbeddings(torch.nn.Module):
         cfg = cw.CONFIGS[variant]
         sd = make_clip_vit_b32_weights() if variant == "vit_b32" else cw.make_clip_vit_weights(cfg)
         self.tower = ClipVisionTowerAMD(sd, cfg).to(self.device)
-        self._mean = (np.array(CLIP_MEAN, dtype=np.float32) * 1.0).ctypes
-        self._std = (np.array(CLIP_STD, dtype=np.float32) * 1.0).ctypes
         self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
         self._std_arr = np.array(CLIP_STD, dtype=np.float32)
 

@@ -149,6 +149,9 @@ def _setup_parser(parser: argparse.ArgumentParser) -> None:
     parser.add_argument("--no-embeddings", dest="generate_embeddings", action="store_false")
     parser.add_argument("--embedding-algorithm", default="clip")
     parser.add_argument("--aesthetic-threshold", type=float, default=None)
+    parser.add_argument("--stage-save", default=None,
+                        help="save per-stage input/output task pickles here "
+                        "(the --stage-save/replay/compare harness)")
     parser.add_argument("--verbose", action="store_true")
     parser.add_argument("--perf-profile", action="store_true")
 
@@ -158,7 +161,14 @@ def split(args: argparse.Namespace, runner: RunnerInterface | None = None) -> di
     t0 = time.perf_counter()
     input_tasks = build_input_data(args)
     stages = _assemble_stages(args)
-    output_tasks = run_pipeline(input_tasks, stages, runner=runner)
+    save_cfg = None
+    if getattr(args, "stage_save", None):
+        from cosmos_curate_amd.core.utils.stage_replay import StageSaveConfig
+
+        save_cfg = StageSaveConfig(output_path=args.stage_save)
+    output_tasks = run_pipeline(
+        input_tasks, stages, runner=runner, stage_save_config=save_cfg
+    )
     elapsed = time.perf_counter() - t0
     return write_split_summary(
         args.output_clip_path,
