@@ -72,7 +72,10 @@ def test_pairwise_kernel_full_size_property():
     argi = argi.cpu().numpy()
     np.testing.assert_allclose(maxv[dup_dst], 1.0, atol=2e-5)
     np.testing.assert_array_equal(argi[dup_dst], dup_src)
-    # random 512-d gaussians are near-orthogonal: non-planted tail rows keep
-    assert (maxv[m - 10_000:][::20][:500] > 0.999).all()
+    # non-planted rows: random 512-d gaussians are near-orthogonal, so
+    # their max cosine stays far below the prune threshold
+    non_planted = m - 10_000 + np.arange(500) * 20 + 1
+    assert (maxv[non_planted] < 0.9).all()
     kept = maxv <= 0.99
     assert not kept[dup_dst].any()
+    assert kept[non_planted].all()
