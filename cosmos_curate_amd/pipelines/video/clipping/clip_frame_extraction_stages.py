@@ -41,6 +41,7 @@ from cosmos_curate_amd.core.interfaces.stage_interface import (
 )
 from cosmos_curate_amd.core.utils.lazy_data import LazyData
 from cosmos_curate_amd.core.utils.performance_utils import StageTimer
+from cosmos_curate_amd.core.utils.roctx import annotate
 from cosmos_curate_amd.pipelines.video.utils import raw_backend
 from cosmos_curate_amd.pipelines.video.utils.data_model import SplitPipeTask, Video
 from cosmos_curate_amd.pipelines.video.utils.decoder_utils import (
@@ -185,6 +186,7 @@ class ClipFrameExtractionStage(CuratorStage):
                 clip.encoded_data.drop()
                 continue
 
+    @annotate("ClipFrameExtractionStage")  # reference: nvtx, clip_frame_extraction_stages.py:167
     def process_data(self, tasks: list[SplitPipeTask]) -> list[SplitPipeTask] | None:
         for task in tasks:
             self._timer.reinit(self, task.get_major_size())

@@ -32,6 +32,7 @@ from cosmos_curate_amd.core.interfaces.stage_interface import (
 from cosmos_curate_amd.core.interfaces.model_interface import ModelInterface
 from cosmos_curate_amd.core.utils.lazy_data import LazyData
 from cosmos_curate_amd.core.utils.performance_utils import StageTimer
+from cosmos_curate_amd.core.utils.roctx import annotate
 from cosmos_curate_amd.models.clip import CLIPImageEmbeddings
 from cosmos_curate_amd.pipelines.video.utils.data_model import Clip, SplitPipeTask
 from cosmos_curate_amd.pipelines.video.utils.decoder_utils import (
@@ -141,6 +142,7 @@ class ClipEmbeddingStage(CuratorStage):
             clip.clip_embedding = e.cpu().numpy().astype(np.float32)
             pos += cnt
 
+    @annotate("ClipEmbeddingStage")
     def process_data(self, tasks: list[SplitPipeTask]) -> list[SplitPipeTask] | None:
         for task in tasks:
             self._timer.reinit(self, task.get_major_size())
