@@ -95,6 +95,10 @@ class Clip:
     clip_embedding_frames: LazyData = dataclasses.field(default_factory=LazyData)
     # aesthetic filter (data_model.py:241)
     aesthetic_score: float | None = None
+    # motion filter (data_model.py:236-239)
+    decoded_motion_data: Any = None
+    motion_score_global_mean: float | None = None
+    motion_score_per_patch_min_256: float | None = None
     errors: dict[str, str] = dataclasses.field(default_factory=dict)
 
     def __post_init__(self) -> None:
