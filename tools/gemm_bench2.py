@@ -20,6 +20,7 @@ from cosmos_curate_amd import hotpath  # noqa: E402
 
 ROOT = pathlib.Path(__file__).resolve().parent.parent
 SO = ROOT / "tools" / "libgemm_variants.so"
+SO8 = ROOT / "tools" / "libgemm_v8.so"
 
 SHAPES = [
     ("patch", 336 * 49, 768, 3072),
@@ -45,7 +46,19 @@ def build() -> ctypes.CDLL:
         ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_int64, ctypes.c_int64, ctypes.c_int64, ctypes.c_int, ctypes.c_uint64,
     ]
-    return lib
+    src8 = ROOT / "tools" / "gemm_v8.hip"
+    if not SO8.exists() or SO8.stat().st_mtime < src8.stat().st_mtime:
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+             f"-I{ROOT}", "-shared", str(src8), "-o", str(SO8)],
+            check=True,
+        )
+    lib8 = ctypes.CDLL(str(SO8))
+    lib8.cc_gemm_v8.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int64, ctypes.c_int64, ctypes.c_int64, ctypes.c_int, ctypes.c_uint64,
+    ]
+    return lib, lib8
 
 
 def time_variant(fn, iters=30) -> float:
@@ -59,9 +72,9 @@ def time_variant(fn, iters=30) -> float:
 
 def main() -> None:
     prod = hotpath.require_gpu()
-    var = build()
+    var, var8 = build()
     stream = torch.cuda.current_stream().cuda_stream
-    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v7':>7s}  TF/s (best of 3 reps)")
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v8':>7s}  TF/s (best of 3 reps)")
     for label, M, N, K in SHAPES:
         torch.manual_seed(1)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
@@ -77,11 +90,16 @@ def main() -> None:
         iters = 20 if M * N * K > 2**36 else 40
         results = []
         calls = {0: prod_call}
-        for v in (2, 6, 7):
+        for v in (2, 6):
             def vcall(v=v):
                 rc = var.cc_gemm_variant(v, a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
                 assert rc == 0
             calls[v] = vcall
+
+        def v8call():
+            rc = var8.cc_gemm_v8(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
+            assert rc == 0
+        calls[8] = v8call
         for v, fn in calls.items():
             if v != 0:
                 c.zero_()
