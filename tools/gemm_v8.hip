@@ -120,8 +120,19 @@ __global__ __launch_bounds__(512, 1) void k_gemm_v8(
           asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
         else
           asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
-      } else if (kt + 1 == KT && q > 0) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      } else if (kt == KT - 2) {
+        // B staging stopped (kt+2 == KT): issue rate 1/phase; the wait
+        // must equal the glds issued in the last 3 phases exactly.
+        if (q == 0)      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        else if (q == 1) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+        else if (q == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else             asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+      } else if (kt == KT - 1) {
+        // A staging stopped too: 0/phase
+        if (q == 0)      asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+        else if (q == 1) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else if (q == 2) asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+        else             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
       }
@@ -185,7 +196,8 @@ __global__ __launch_bounds__(512, 1) void k_gemm_v8(
 
 extern "C" int cc_gemm_v8(const void* A, const void* B, void* C, int64_t M,
                           int64_t N, int64_t K, int c_dtype, uint64_t stream) {
-  if (K % 64 != 0) return cc::set_error(CC_ERR_UNSUPPORTED, "K%%64");
+  if (K % 64 != 0 || K / 64 < 5)
+    return cc::set_error(CC_ERR_UNSUPPORTED, "v8 needs K%%64==0 and K>=320");
   dim3 block(512);
   dim3 grid((N + 255) / 256, (M + 255) / 256);
   hipLaunchKernelGGL(v8::k_gemm_v8, grid, block, 0, (hipStream_t)stream,
