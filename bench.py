@@ -109,7 +109,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--clips", type=int, default=32, help="clips per step per rank")
+    ap.add_argument("--clips", type=int, default=64, help="clips per step per rank")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument(
         "--graphs", action="store_true", default=True,
