@@ -224,3 +224,29 @@ def test_clip_model_interface_end_to_end(lib):
     want = oracle_vit.embed_frames_fp32(ref, clip_preprocess(frames))
     cos = np.sum(want * emb.cpu().numpy(), axis=1)
     assert np.all(cos >= 0.999), cos
+
+
+def test_iv2_preprocess_matches_oracle(lib):
+    """IV2 input-frame math (subsample/resize/normalize) — SURVEY.md §2
+    'Embedding models' preprocessing-parity contract."""
+    from cosmos_curate_amd.models import internvideo2_prep as prep
+    from oracle import iv2_preprocess as oprep
+
+    rng = np.random.default_rng(0x12)
+    frames = rng.integers(0, 256, size=(21, 256, 320, 3), dtype=np.uint8)
+    want = oprep.formulate_input_frames(frames, fnum=8, target_size=224)
+    got = prep.formulate_input_frames(
+        torch.from_numpy(frames).cuda(), fnum=8, target_size=224
+    ).cpu().numpy()
+    assert got.shape == want.shape == (1, 8, 3, 224, 224)
+    np.testing.assert_array_equal(got, want)  # bit-exact: same f32 chain
+
+    # subsample rule parity at awkward lengths
+    from cosmos_curate_amd.models.internvideo2_prep import temporal_subsample
+
+    for t in [8, 9, 15, 21, 100]:
+        np.testing.assert_array_equal(
+            temporal_subsample(t, 8), np.arange(t)[:: t // 8][:8]
+        )
+    with pytest.raises(ValueError):
+        temporal_subsample(5, 8)
