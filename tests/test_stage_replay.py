@@ -69,3 +69,34 @@ def test_replay_stage_runs_setup(tmp_path):
                  runner=SequentialRunner(), stage_save_config=cfg)
     outs = replay_stage(FixedStrideExtractorStage(), str(tmp_path))
     assert len(outs) == 1 and len(outs[0][0].video.clips) == 3
+
+
+def test_profiling_wrapper_cpu_backend(tmp_path):
+    from cosmos_curate_amd.core.utils.profiling import (
+        ProfilingConfig,
+        profiling_wrapper,
+    )
+
+    stage = profiling_wrapper(
+        FixedStrideExtractorStage(),
+        ProfilingConfig(output_path=str(tmp_path), profile_cpu=True),
+    )
+    out = run_pipeline([make_task()], [stage], runner=SequentialRunner())
+    assert len(out[0].video.clips) == 3
+    dumps = list((tmp_path / "profile" / "FixedStrideExtractorStage").glob("cpu_*.pstats"))
+    assert len(dumps) == 1
+    import pstats
+
+    st = pstats.Stats(str(dumps[0]))
+    assert st.total_calls > 0
+
+
+def test_profiling_wrapper_disabled_is_identity(tmp_path):
+    from cosmos_curate_amd.core.utils.profiling import (
+        ProfilingConfig,
+        profiling_wrapper,
+    )
+
+    s = FixedStrideExtractorStage()
+    assert profiling_wrapper(s, None) is s
+    assert profiling_wrapper(s, ProfilingConfig(output_path=str(tmp_path))) is s
