@@ -1,0 +1,83 @@
+"""Sharding pipeline: bin spec, greedy packer, tar round-trip (CPU)."""
+
+import json
+import tarfile
+
+import numpy as np
+import pytest
+
+from cosmos_curate_amd.pipelines.video import sharding_pipeline as shp
+
+
+def sample(uuid, nbytes=1000, w=1920, h=1080, fps=30.0, frames=300):
+    return shp.ClipSample(uuid=uuid, payload=bytes(nbytes), width=w, height=h,
+                          framerate=fps, num_frames=frames,
+                          metadata={"uuid": uuid, "n": nbytes})
+
+
+def test_bin_key_classes():
+    assert shp.bin_key(sample("a")) == "res1080_ar16-9_5-10s"
+    assert shp.bin_key(sample("b", w=640, h=480, frames=60)) == "res480_ar4-3_0-2s"
+    assert shp.bin_key(sample("c", w=1080, h=1920)) == "res1080_ar9-16_5-10s"
+    assert shp.bin_key(sample("d", fps=0.0)) is None
+
+
+def test_greedy_packer_semantics():
+    ss = [sample(f"s{i}", nbytes=400) for i in range(7)]
+    packs = list(shp.group_samples_by_size(ss, 1000))
+    assert [len(p) for p in packs] == [2, 2, 2, 1]
+    # small tail dropped when requested (min_clips_per_tar=2)
+    packs = list(shp.group_samples_by_size(ss, 1000, drop_small_shards=True))
+    assert [len(p) for p in packs] == [2, 2, 2]
+    # oversize sample still gets its own tar
+    packs = list(shp.group_samples_by_size([sample("big", nbytes=5000)], 1000))
+    assert [len(p) for p in packs] == [1]
+
+
+def test_write_and_roundtrip(tmp_path):
+    ss = [sample(f"u{i}", nbytes=500 + i) for i in range(5)]
+    ss.append(sample("v0", w=640, h=480))
+    index = shp.write_webdataset_shards(ss, str(tmp_path), target_size_bytes=1200)
+    disk = json.loads((tmp_path / "shard_index.json").read_text())
+    assert disk == index
+    assert "res1080_ar16-9_5-10s" in index and "res480_ar4-3_5-10s" in index
+    total = 0
+    for label, shards in index.items():
+        for s in shards:
+            with tarfile.open(tmp_path / s["tar"]) as tar:
+                names = tar.getnames()
+                bins = [n for n in names if n.endswith(".bin")]
+                metas = [n for n in names if n.endswith(".json")]
+                assert len(bins) == len(metas) == s["clips"]
+                for m in metas:
+                    meta = json.loads(tar.extractfile(m).read())
+                    payload = tar.extractfile(meta["uuid"] + ".bin").read()
+                    assert len(payload) == meta["n"]
+            total += s["clips"]
+    assert total == 6
+
+
+def test_shard_from_split_output(tmp_path):
+    """split (raw corpus, no embeddings) -> shard end-to-end on CPU."""
+    import argparse
+
+    from cosmos_curate_amd.core.interfaces import SequentialRunner
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import _setup_parser, split
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    inp = tmp_path / "in"
+    inp.mkdir()
+    (inp / "v.nv12").write_bytes(raw_backend.make_synthetic_clip(600, 64, 96, 30, seed=1))
+    out = tmp_path / "split"
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    args = p.parse_args(["--input-video-path", str(inp), "--output-clip-path",
+                         str(out), "--no-embeddings"])
+    split(args, runner=SequentialRunner())
+
+    index = shp.cli_run_shard([
+        "--input-clip-path", str(out), "--output-shard-path", str(tmp_path / "shards"),
+        "--target-shard-size-mb", "1",
+    ])
+    total = sum(s["clips"] for shards in index.values() for s in shards)
+    assert total == 2  # 20 s video -> two 10 s clips
