@@ -393,6 +393,8 @@ int cc_nv12_to_rgb(const void* y, const void* uv, int n, int h, int w,
                    size_t pitch, void* out_rgb, uint64_t stream) {
   if (!y || !uv || !out_rgb || n <= 0 || h <= 0 || w <= 0 || pitch < (size_t)w)
     return cc::set_error(CC_ERR_INVALID, "bad nv12 args");
+  if ((h & 1) || (w & 1))
+    return cc::set_error(CC_ERR_INVALID, "NV12 needs even dims (got %dx%d)", w, h);
   long total = (long)n * h * w;
   dim3 block(256), grid((total + 255) / 256);
   CC_LAUNCH("nv12_to_rgb", grid, block, stream, k_nv12_to_rgb,
@@ -407,6 +409,9 @@ int cc_nv12_to_rgb_resize(const void* y, const void* uv, int n, int src_h,
   if (!y || !uv || !out_rgb || n <= 0 || src_h <= 0 || src_w <= 0 ||
       out_h <= 0 || out_w <= 0 || pitch < (size_t)src_w)
     return cc::set_error(CC_ERR_INVALID, "bad nv12_resize args");
+  if ((src_h & 1) || (src_w & 1))
+    return cc::set_error(CC_ERR_INVALID, "NV12 needs even dims (got %dx%d)",
+                         src_w, src_h);
   long total = (long)n * out_h * out_w;
   dim3 block(256), grid((total + 255) / 256);
   CC_LAUNCH("nv12_to_rgb_resize", grid, block, stream, k_nv12_to_rgb_resize,

@@ -131,3 +131,14 @@ def test_gather_broadcast_matches_oracle(lib):
     torch.cuda.synchronize()
     want = osampling.broadcast_selected(frames, idx, counts)
     np.testing.assert_array_equal(out.cpu().numpy(), want)
+
+
+def test_nv12_rejects_odd_dims(lib):
+    buf = torch.zeros(64 * 63, dtype=torch.uint8, device="cuda")
+    out = torch.zeros(64 * 64 * 3, dtype=torch.uint8, device="cuda")
+    rc = lib.cc_nv12_to_rgb(buf.data_ptr(), buf.data_ptr(), 1, 63, 64, 64,
+                            out.data_ptr(), 0)
+    assert rc != 0
+    rc = lib.cc_nv12_to_rgb_resize(buf.data_ptr(), buf.data_ptr(), 1, 64, 63, 64,
+                                   out.data_ptr(), 32, 32, 0)
+    assert rc != 0
