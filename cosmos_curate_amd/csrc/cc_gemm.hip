@@ -188,7 +188,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
         for (int r = 0; r < 4; r++) {
           const long row = crow_base + m * FRAG + r;
           float v = acc[m][n][r] + bval;
-          if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+          if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
           if constexpr (HAS_RES) v += rv[r];
           if (c_is_bf16)
             ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -212,7 +212,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
         const long row = crow_base + m * FRAG + r;
         if (row >= M) continue;
         float v = acc[m][n][r] + bval;
-        if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));  // quick-gelu
+        if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));  // quick-gelu
         if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -308,7 +308,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
         for (int reg = 0; reg < 16; reg++) {
           const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
           float v = acc[m][n][reg] + bval;
-          if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+          if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
           if constexpr (HAS_RES) v += rv[reg];
           if (c_is_bf16)
             ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -332,7 +332,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
         const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
         if (row >= M) continue;
         float v = acc[m][n][reg] + bval;
-        if (ACT == 1) v = v / (1.0f + __expf(-1.702f * v));
+        if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
         if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
