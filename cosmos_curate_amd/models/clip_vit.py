@@ -21,6 +21,7 @@ product never does).
 
 from __future__ import annotations
 
+import ctypes
 import math
 
 import torch
@@ -198,8 +199,8 @@ class ClipVisionTowerAMD(torch.nn.Module):
                 hotpath.check(
                     lib.cc_attn_small(
                         qkv_flat.data_ptr(), attn.data_ptr(), n, seq,
-                        self.heads, cfg.hidden,
-                        __import__("ctypes").c_float(self.scale), stream,
+                        self.heads, cfg.hidden, ctypes.c_float(self.scale),
+                        stream,
                     )
                 )
             else:

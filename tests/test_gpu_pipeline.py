@@ -123,9 +123,9 @@ def test_mp4_clip_without_rocdecode_fails_loudly():
     decode_unavailable, never silently CPU-decode."""
     task, _ = make_task(1)
     golden = pathlib.Path(__file__).parent / "golden" / "synth_bframes.mp4"
-    task.video.clips[0].encoded_data = __import__(
-        "cosmos_curate_amd.core.utils.lazy_data", fromlist=["LazyData"]
-    ).LazyData.coerce(golden.read_bytes())
+    from cosmos_curate_amd.core.utils.lazy_data import LazyData
+
+    task.video.clips[0].encoded_data = LazyData.coerce(golden.read_bytes())
     stage = ClipFrameExtractionStage(target_fps=[2], target_res=(224, 224))
     out = run_pipeline([task], [stage], runner=SequentialRunner())
     clip = out[0].video.clips[0]
