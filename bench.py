@@ -119,6 +119,11 @@ def main() -> None:
     )
     ap.add_argument("--no-graphs", dest="graphs", action="store_false")
     ap.add_argument(
+        "--streams", type=int, default=1, choices=[1, 2],
+        help="split each step's batch across N HIP streams (overlaps one "
+        "half's GEMM chain with the other's small kernels)",
+    )
+    ap.add_argument(
         "--model", default="vit_b32", choices=["vit_b32", "vit_l14"],
         help="embedder tower: vit_b32 (flagship, configs #1/#2) or vit_l14 "
         "(the reference's CLIP model / config #3 class)",
@@ -152,6 +157,8 @@ def main() -> None:
     rgb = torch.empty((F, RES, RES, 3), dtype=torch.uint8, device=device)
     stream = torch.cuda.current_stream(device).cuda_stream
 
+    side_streams = [torch.cuda.Stream(device) for _ in range(args.streams - 1)]
+
     def step_device() -> torch.Tensor:
         hotpath.check(
             lib.cc_nv12_to_rgb_resize(
@@ -159,8 +166,24 @@ def main() -> None:
                 rgb.data_ptr(), RES, RES, stream,
             )
         )
-        pixels = model.preprocess_u8(rgb)
-        emb = model.tower(pixels)  # (F, 512) f32 unit-norm
+        if args.streams == 1:
+            pixels = model.preprocess_u8(rgb)
+            emb = model.tower(pixels)  # (F, proj) f32 unit-norm
+        else:
+            # split the batch across streams: one half's bandwidth-bound
+            # kernels overlap the other half's MFMA chain
+            main = torch.cuda.current_stream(device)
+            half = F // 2
+            parts = []
+            chunks = [rgb[:half], rgb[half:]]
+            for st, chunk in zip([main, *side_streams], chunks):
+                with torch.cuda.stream(st):
+                    if st is not main:
+                        st.wait_stream(main)
+                    parts.append(model.tower(model.preprocess_u8(chunk)))
+            for st in side_streams:
+                main.wait_stream(st)
+            emb = torch.cat(parts, dim=0)
         per_clip = emb.view(B, FRAMES_PER_CLIP, emb.shape[-1]).mean(dim=1)
         return per_clip / torch.linalg.vector_norm(per_clip, dim=-1, keepdim=True)
 
