@@ -23,11 +23,9 @@ the world_size-2 gloo test drives the collective path on CPU.
 
 from __future__ import annotations
 
-import ctypes
 import dataclasses
 
 import numpy as np
-import numpy.typing as npt
 import torch
 
 from cosmos_curate_amd import hotpath

@@ -6,7 +6,6 @@
   committed fixtures and on randomized writer-generated containers.
 """
 
-import ctypes
 import json
 import re
 import pathlib

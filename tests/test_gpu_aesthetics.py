@@ -5,7 +5,6 @@ import uuid
 
 import numpy as np
 import pytest
-import torch
 
 from cosmos_curate_amd.core.interfaces import SequentialRunner, run_pipeline
 from cosmos_curate_amd.core.utils.lazy_data import LazyData

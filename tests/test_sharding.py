@@ -4,7 +4,6 @@ import json
 import tarfile
 
 import numpy as np
-import pytest
 
 from cosmos_curate_amd.pipelines.video import sharding_pipeline as shp
 
