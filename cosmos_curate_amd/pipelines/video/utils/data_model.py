@@ -274,3 +274,56 @@ class SplitPipeTask(PipelineTask):
 
     def get_major_size(self) -> int:
         return get_major_size(self)
+
+
+def check_clip_time_alignment(clips_per_video: list[list[Clip]]) -> list[int]:
+    """Multicam: indices where same-index clips have differing spans
+    (data_model.py:595-631).  Raises when clip counts differ."""
+    if not clips_per_video:
+        return []
+    counts = [len(c) for c in clips_per_video]
+    if not all(c == counts[0] for c in counts):
+        msg = (
+            f"Cannot check time alignment: videos have different clip counts "
+            f"{counts}. All videos must have the same number of clips."
+        )
+        raise ValueError(msg)
+    misaligned = []
+    for i in range(counts[0]):
+        spans = [clips[i].span for clips in clips_per_video]
+        if not all(s == spans[0] for s in spans):
+            misaligned.append(i)
+    return misaligned
+
+
+def assert_video_clip_alignment(videos: list[Video]) -> None:
+    """Multicam sync validation (data_model.py:634-687): same processed
+    count per camera, identical spans per index across clips AND
+    filtered_clips."""
+    if not videos:
+        return
+    processed = [len(v.clips) + len(v.filtered_clips) for v in videos]
+    if not all(p == processed[0] for p in processed):
+        msg = (
+            f"Multi-cam videos have processed different numbers of clips: "
+            f"{processed}."
+        )
+        raise ValueError(msg)
+    for field in ("clips", "filtered_clips"):
+        groups = [getattr(v, field) for v in videos]
+        bad = check_clip_time_alignment(groups)
+        if bad:
+            spans = [getattr(v, field)[bad[0]].span for v in videos]
+            msg = (
+                f"Multi-cam {field} at index {bad[0]} have misaligned spans: "
+                f"{spans}. Misaligned indices: {bad}"
+            )
+            raise ValueError(msg)
+
+
+def assert_time_alignment(tasks: list["SplitPipeTask"]) -> None:
+    """Per-task multicam validation over a chunked task list
+    (data_model.py:862-877; consumed by chunk_tasks)."""
+    for task in tasks:
+        if len(task.videos) > 1:
+            assert_video_clip_alignment(task.videos)

@@ -139,3 +139,53 @@ def test_retry_attempts():
     spec = CuratorStageSpec(Flaky(), num_run_attempts_python=3)
     out = run_pipeline([make_task()], [spec], runner=SequentialRunner())
     assert calls["n"] == 3 and len(out) == 1
+
+
+def test_multicam_alignment_checks():
+    """data_model.py:595-687 multicam contract: aligned spans pass,
+    misalignment and count mismatch are loud."""
+    import uuid as uuid_mod
+
+    from cosmos_curate_amd.pipelines.video.utils.data_model import (
+        Clip,
+        assert_video_clip_alignment,
+        check_clip_time_alignment,
+    )
+
+    def cam(spans):
+        v = make_task().video
+        v.clips = [
+            Clip(uuid=uuid_mod.uuid4(), source_video="s", span=s) for s in spans
+        ]
+        return v
+
+    a = cam([(0.0, 10.0), (10.0, 20.0)])
+    b = cam([(0.0, 10.0), (10.0, 20.0)])
+    assert check_clip_time_alignment([a.clips, b.clips]) == []
+    assert_video_clip_alignment([a, b])  # no raise
+
+    c = cam([(0.0, 10.0), (10.0, 21.0)])
+    assert check_clip_time_alignment([a.clips, c.clips]) == [1]
+    with pytest.raises(ValueError, match="misaligned spans"):
+        assert_video_clip_alignment([a, c])
+
+    d = cam([(0.0, 10.0)])
+    with pytest.raises(ValueError, match="different clip counts"):
+        check_clip_time_alignment([a.clips, d.clips])
+
+
+def test_multicam_fixed_stride_shares_spans_and_uuids():
+    """Multicam: FixedStrideExtractorStage gives every camera the same
+    spans AND the same clip uuids (clip_extraction_stages.py:654-661)."""
+    from cosmos_curate_amd.pipelines.video.utils.data_model import (
+        assert_time_alignment,
+    )
+
+    t = make_task()
+    t.videos.append(make_task().video)  # second camera
+    out = run_pipeline([t], [FixedStrideExtractorStage()], runner=SequentialRunner())
+    cams = out[0].videos
+    assert len(cams) == 2
+    assert [c.span for c in cams[0].clips] == [c.span for c in cams[1].clips]
+    assert [c.uuid for c in cams[0].clips] == [c.uuid for c in cams[1].clips]
+    assert_time_alignment(out)
