@@ -212,6 +212,12 @@ def chunk_tasks(
             )
             out.append(sub)
             start = end
+    # multicam chunking must preserve time alignment (reference :163)
+    from cosmos_curate_amd.pipelines.video.utils.data_model import (
+        assert_time_alignment,
+    )
+
+    assert_time_alignment(out)
     return out
 
 
