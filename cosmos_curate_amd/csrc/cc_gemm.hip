@@ -30,6 +30,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <vector>
 
 #include "cc_common.hpp"
@@ -366,7 +367,12 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   bool timed = cc::timed_begin(stream, &ev0, &ev1);
   // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
   // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
-  const bool wide = K < 2048;
+  // CC_GEMM_WIDE=0|1 overrides for A/B experiments.
+  static const int wide_env = [] {
+    const char* e = getenv("CC_GEMM_WIDE");
+    return e ? atoi(e) : -1;
+  }();
+  const bool wide = wide_env >= 0 ? wide_env != 0 : K < 2048;
   const bool hb = bias != nullptr;
   const bool hr = residual != nullptr;
   // XCD remap only for outputs that spill L3 (measured negative on the
