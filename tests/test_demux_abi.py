@@ -128,3 +128,29 @@ def test_product_metadata_from_demux(built_lib, golden_dir):
     assert 29.0 < md.fps < 31.0
     ts = get_video_timestamps(data)
     assert len(ts) == 24
+
+
+def test_demux_parity_multi_entry_stts(built_lib):
+    """VFR-style multi-entry stts (+ large elst offsets): C++ == oracle."""
+    rng = np.random.default_rng(0xBEEF)
+    for trial in range(12):
+        ts = int(rng.choice([12288, 90000, 1000, 600]))
+        entries = []
+        n = 0
+        for _ in range(int(rng.integers(2, 6))):
+            cnt = int(rng.integers(1, 12))
+            entries.append((cnt, int(rng.integers(50, 5000))))
+            n += cnt
+        # edit-list media_time several sample durations in (beyond one GOP)
+        elst = int(entries[0][1] * rng.integers(0, 5)) if trial % 2 else None
+        sizes = [int(rng.integers(8, 200)) for _ in range(n)]
+        data = mp4_write.write_mp4(
+            sizes, stts=entries, ctts=None, timescale=ts,
+            elst_media_time=elst, sync_samples=[1],
+        )
+        oracle_ts = mp4_demux.get_video_timestamps(data)
+        with hotpath.Demuxer(data) as d:
+            got = d.timestamps()
+            info = d.probe()
+        np.testing.assert_array_equal(got, oracle_ts)
+        assert info.num_samples == n
