@@ -74,7 +74,7 @@ def main() -> None:
     prod = hotpath.require_gpu()
     var, var8 = build()
     stream = torch.cuda.current_stream().cuda_stream
-    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v8':>7s}  TF/s (best of 3 reps)")
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v9':>7s} {'v8':>7s}  TF/s (best of 3 reps)")
     for label, M, N, K in SHAPES:
         torch.manual_seed(1)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
@@ -90,7 +90,7 @@ def main() -> None:
         iters = 20 if M * N * K > 2**36 else 40
         results = []
         calls = {0: prod_call}
-        for v in (2, 6):
+        for v in (2, 6, 9):
             def vcall(v=v):
                 rc = var.cc_gemm_variant(v, a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
                 assert rc == 0

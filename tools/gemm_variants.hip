@@ -464,6 +464,100 @@ __global__ __launch_bounds__(256, 2) void k_gemm_v7(
 }
 }  // namespace v7
 
+
+// ---- V9: 16x16x32, register-double-buffered fragments + counted waits ----
+// The production body's asm shows 4x full `s_waitcnt lgkmcnt(0)` per
+// K-phase: the compiler reused 24 staging VGPRs, serializing each 8-MFMA
+// group on a full ds_read drain.  Here both k-groups (16 frags, 64 VGPRs)
+// are read up front, ds_reads issued BEFORE the glds m0 chain, so the
+// waitcnt pass can emit counted waits (lgkmcnt(8)) and the second MFMA
+// group covers the remaining latency.
+namespace v9 {
+constexpr int WM = 64, WN = 64, FRAG = 16;
+constexpr int MFR = WM / FRAG, NFR = WN / FRAG;
+
+__global__ __launch_bounds__(256, 2) void k_gemm_v9(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];
+#define A9(b) (lds + (b) * (BM * BK))
+#define B9(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 1, waveN = wid & 1;
+  const long bm = (long)blockIdx.y * BM;
+  const long bn = (long)blockIdx.x * BN;
+  const long arow0 = bm + 32 * wid;
+  const long brow0 = bn + 32 * wid;
+
+  f32x4 acc[MFR][NFR] = {};
+  const long KT = K / BK;
+  stage_slice(A, K, arow0, M, 0, A9(0) + 32 * wid * BK, lane);
+  stage_slice(B, K, brow0, N, 0, B9(0) + 32 * wid * BK, lane);
+
+  const int arow_frag = waveM * WM + (lane & 15);
+  const int brow_frag = waveN * WN + (lane & 15);
+  int buf = 0;
+  for (long kt = 0; kt < KT; ++kt) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    const __bf16* At = A9(buf);
+    const __bf16* Bt = B9(buf);
+    // both k-groups up front: 16 ds_read_b128 into 64 live VGPRs
+    bf16x8 afrag[2][MFR], bfrag[2][NFR];
+#pragma unroll
+    for (int g = 0; g < 2; g++) {
+      const int k16 = (g << 2) + (lane >> 4);
+#pragma unroll
+      for (int m = 0; m < MFR; m++)
+        afrag[g][m] = frag_read(At, arow_frag + m * FRAG, k16);
+#pragma unroll
+      for (int n = 0; n < NFR; n++)
+        bfrag[g][n] = frag_read(Bt, brow_frag + n * FRAG, k16);
+    }
+    // glds for the next tile issued after the reads (vmcnt domain; the
+    // m0/readfirstlane chain now runs under the ds_read latency)
+    if (kt + 1 < KT) {
+      const long k0 = (kt + 1) * BK;
+      stage_slice(A, K, arow0, M, k0, A9(buf ^ 1) + 32 * wid * BK, lane);
+      stage_slice(B, K, brow0, N, k0, B9(buf ^ 1) + 32 * wid * BK, lane);
+    }
+#pragma unroll
+    for (int g = 0; g < 2; g++)
+#pragma unroll
+      for (int m = 0; m < MFR; m++)
+#pragma unroll
+        for (int n = 0; n < NFR; n++)
+          acc[g == 0 ? m : m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[g][m], bfrag[g][n], acc[m][n], 0, 0, 0);
+    buf ^= 1;
+  }
+  const long crow_base = bm + waveM * WM + 4 * (lane >> 4);
+  const long ccol_base = bn + waveN * WN + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < MFR; m++) {
+#pragma unroll
+    for (int n = 0; n < NFR; n++) {
+      const long col = ccol_base + n * FRAG;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const long row = crow_base + m * FRAG + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+  }
+#undef A9
+#undef B9
+}
+}  // namespace v9
+
 extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* C,
                                int64_t M, int64_t N, int64_t K, int c_dtype,
                                uint64_t stream) {
@@ -497,8 +591,12 @@ extern "C" int cc_gemm_variant(int variant, const void* A, const void* B, void* 
     hipLaunchKernelGGL(v7::k_gemm_v7, grid, block, 0, (hipStream_t)stream,
                        (const __bf16*)A, (const __bf16*)B, C, (long)M,
                        (long)N, (long)K, c_dtype == 1);
+  } else if (variant == 9) {
+    hipLaunchKernelGGL(v9::k_gemm_v9, grid, block, 0, (hipStream_t)stream,
+                       (const __bf16*)A, (const __bf16*)B, C, (long)M,
+                       (long)N, (long)K, c_dtype == 1);
   } else
-    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4|5|6|7");
+    return cc::set_error(CC_ERR_INVALID, "variant must be 2|3|4|5|6|7|9");
   hipError_t e = hipGetLastError();
   if (e != hipSuccess) return cc::set_error(CC_ERR_HIP, "%s", hipGetErrorString(e));
   return CC_OK;
