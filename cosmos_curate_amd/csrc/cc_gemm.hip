@@ -129,36 +129,44 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
   stage_slice(A, K, arow0, M, 0, AS(0) + 32 * wid * BK, lane);
   stage_slice(B, K, brow0, N, 0, BS(0) + 32 * wid * BK, lane);
 
+  const int arow_frag = waveM * WM + (lane & 15);
+  const int brow_frag = waveN * WN + (lane & 15);
   int buf = 0;
   for (long kt = 0; kt < KT; ++kt) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
+    const __bf16* At = AS(buf);
+    const __bf16* Bt = BS(buf);
+    // Register-double-buffered fragments (graduated v9, profiles/
+    // r01_gemm_v9.log): read BOTH k-groups up front (16 ds_read_b128 into
+    // 64 live VGPRs) and issue the glds m0 chain under their latency, so
+    // the phase runs ONE wait + 32 back-to-back MFMAs.  The previous
+    // per-group form made hipcc reuse 24 staging VGPRs = 4 full
+    // lgkmcnt(0) drains per phase (+12-18% measured from this change).
+    bf16x8 afrag[2][MFR], bfrag[2][NFR];
+#pragma unroll
+    for (int g = 0; g < 2; g++) {
+      const int k16 = (g << 2) + (lane >> 4);
+#pragma unroll
+      for (int m = 0; m < MFR; m++)
+        afrag[g][m] = frag_read(At, arow_frag + m * FRAG, k16);
+#pragma unroll
+      for (int n = 0; n < NFR; n++)
+        bfrag[g][n] = frag_read(Bt, brow_frag + n * FRAG, k16);
+    }
     if (kt + 1 < KT) {
       const long k0 = (kt + 1) * BK;
       stage_slice(A, K, arow0, M, k0, AS(buf ^ 1) + 32 * wid * BK, lane);
       stage_slice(B, K, brow0, N, k0, BS(buf ^ 1) + 32 * wid * BK, lane);
     }
-    const __bf16* At = AS(buf);
-    const __bf16* Bt = BS(buf);
-    const int arow_frag = waveM * WM + (lane & 15);
-    const int brow_frag = waveN * WN + (lane & 15);
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 32) {
-      const int k16 = (kk >> 3) + (lane >> 4);
-      bf16x8 afrag[MFR], bfrag[NFR];
-#pragma unroll
-      for (int m = 0; m < MFR; m++)
-        afrag[m] = frag_read(At, arow_frag + m * FRAG, k16);
-#pragma unroll
-      for (int n = 0; n < NFR; n++)
-        bfrag[n] = frag_read(Bt, brow_frag + n * FRAG, k16);
+    for (int g = 0; g < 2; g++)
 #pragma unroll
       for (int m = 0; m < MFR; m++)
 #pragma unroll
         for (int n = 0; n < NFR; n++)
           acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag[m], bfrag[n], acc[m][n], 0, 0, 0);
-    }
+              afrag[g][m], bfrag[g][n], acc[m][n], 0, 0, 0);
     buf ^= 1;
   }
 
@@ -258,31 +266,38 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
   const long KT = K / BK;
   stage_slice(A, K, arow0, M, 0, AS32(0) + 32 * wid * BK, lane);
   stage_slice(B, K, brow0, N, 0, BS32(0) + 32 * wid * BK, lane);
+  const int arow = waveM * WM + (lane & 31);
+  const int brow = waveN * WN + (lane & 31);
   int buf = 0;
   for (long kt = 0; kt < KT; ++kt) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
+    const __bf16* At = AS32(buf);
+    const __bf16* Bt = BS32(buf);
+    // register-double-buffered fragments across all 4 k-steps (same v9
+    // restructure as the 16x16 body above: one wait per phase, glds m0
+    // chain issued under the ds_read latency)
+    bf16x8 a0[4], a1[4], b0[4], b1[4];
+#pragma unroll
+    for (int g = 0; g < 4; g++) {
+      // 32x32x16 A/B fragment: lane l holds row l&31, k 8*(l>>5)..+8
+      const int k16 = (g << 1) + (lane >> 5);
+      a0[g] = frag_read(At, arow, k16);
+      a1[g] = frag_read(At, arow + 32, k16);
+      b0[g] = frag_read(Bt, brow, k16);
+      b1[g] = frag_read(Bt, brow + 32, k16);
+    }
     if (kt + 1 < KT) {
       const long k0 = (kt + 1) * BK;
       stage_slice(A, K, arow0, M, k0, AS32(buf ^ 1) + 32 * wid * BK, lane);
       stage_slice(B, K, brow0, N, k0, BS32(buf ^ 1) + 32 * wid * BK, lane);
     }
-    const __bf16* At = AS32(buf);
-    const __bf16* Bt = BS32(buf);
-    const int arow = waveM * WM + (lane & 31);
-    const int brow = waveN * WN + (lane & 31);
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 16) {
-      // 32x32x16 A/B fragment: lane l holds row l&31, k 8*(l>>5)..+8
-      const int k16 = (kk >> 3) + (lane >> 5);
-      bf16x8 a0 = frag_read(At, arow, k16);
-      bf16x8 a1 = frag_read(At, arow + 32, k16);
-      bf16x8 b0 = frag_read(Bt, brow, k16);
-      bf16x8 b1 = frag_read(Bt, brow + 32, k16);
-      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    for (int g = 0; g < 4; g++) {
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[g], b0[g], acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[g], b1[g], acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[g], b0[g], acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[g], b1[g], acc[1][1], 0, 0, 0);
     }
     buf ^= 1;
   }
