@@ -269,35 +269,32 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
   const int arow = waveM * WM + (lane & 31);
   const int brow = waveN * WN + (lane & 31);
   int buf = 0;
+  // NOTE: the v9 register-double-buffer restructure (16x16 body above)
+  // was tried here too and measured SLOWER (qkv 570 vs 582 TF,
+  // profiles/r01_v9_graduated.log) — the w32 body keeps the per-k-step
+  // form, and the default dispatch no longer selects it (see launcher).
   for (long kt = 0; kt < KT; ++kt) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    const __bf16* At = AS32(buf);
-    const __bf16* Bt = BS32(buf);
-    // register-double-buffered fragments across all 4 k-steps (same v9
-    // restructure as the 16x16 body above: one wait per phase, glds m0
-    // chain issued under the ds_read latency)
-    bf16x8 a0[4], a1[4], b0[4], b1[4];
-#pragma unroll
-    for (int g = 0; g < 4; g++) {
-      // 32x32x16 A/B fragment: lane l holds row l&31, k 8*(l>>5)..+8
-      const int k16 = (g << 1) + (lane >> 5);
-      a0[g] = frag_read(At, arow, k16);
-      a1[g] = frag_read(At, arow + 32, k16);
-      b0[g] = frag_read(Bt, brow, k16);
-      b1[g] = frag_read(Bt, brow + 32, k16);
-    }
     if (kt + 1 < KT) {
       const long k0 = (kt + 1) * BK;
       stage_slice(A, K, arow0, M, k0, AS32(buf ^ 1) + 32 * wid * BK, lane);
       stage_slice(B, K, brow0, N, k0, BS32(buf ^ 1) + 32 * wid * BK, lane);
     }
+    const __bf16* At = AS32(buf);
+    const __bf16* Bt = BS32(buf);
 #pragma unroll
-    for (int g = 0; g < 4; g++) {
-      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[g], b0[g], acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[g], b1[g], acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[g], b0[g], acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[g], b1[g], acc[1][1], 0, 0, 0);
+    for (int kk = 0; kk < BK; kk += 16) {
+      // 32x32x16 A/B fragment: lane l holds row l&31, k 8*(l>>5)..+8
+      const int k16 = (kk >> 3) + (lane >> 5);
+      bf16x8 a0 = frag_read(At, arow, k16);
+      bf16x8 a1 = frag_read(At, arow + 32, k16);
+      bf16x8 b0 = frag_read(Bt, brow, k16);
+      bf16x8 b1 = frag_read(Bt, brow + 32, k16);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
     }
     buf ^= 1;
   }
@@ -380,14 +377,16 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   dim3 grid(nwg);
   hipEvent_t ev0 = nullptr, ev1 = nullptr;
   bool timed = cc::timed_begin(stream, &ev0, &ev1);
-  // measured dispatch rule (profiles/r01_gemm_variants): the 32x32x16 body
-  // wins below K=2048, the 16x16x32 body at/above (patch-embed K=3072).
-  // CC_GEMM_WIDE=0|1 overrides for A/B experiments.
+  // After the v9 register-double-buffer restructure the 16x16x32 body
+  // wins at EVERY measured shape (profiles/r01_v9_graduated.log: qkv
+  // 671 vs 570, squares 1051/1059 vs 1015/897 TF), so it is the default
+  // everywhere; the old K<2048 rule picked the 32x32x16 body.
+  // CC_GEMM_WIDE=1 re-enables the w32 body for A/B experiments.
   static const int wide_env = [] {
     const char* e = getenv("CC_GEMM_WIDE");
     return e ? atoi(e) : -1;
   }();
-  const bool wide = wide_env >= 0 ? wide_env != 0 : K < 2048;
+  const bool wide = wide_env >= 0 ? wide_env != 0 : false;
   const bool hb = bias != nullptr;
   const bool hr = residual != nullptr;
   // XCD remap only for outputs that spill L3 (measured negative on the
