@@ -58,7 +58,17 @@ def build() -> ctypes.CDLL:
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_int64, ctypes.c_int64, ctypes.c_int64, ctypes.c_int, ctypes.c_uint64,
     ]
-    return lib, lib8
+    src10 = ROOT / "tools" / "gemm_v10.hip"
+    SO10 = ROOT / "tools" / "libgemm_v10.so"
+    if not SO10.exists() or SO10.stat().st_mtime < src10.stat().st_mtime:
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+             f"-I{ROOT}", "-shared", str(src10), "-o", str(SO10)],
+            check=True,
+        )
+    lib10 = ctypes.CDLL(str(SO10))
+    lib10.cc_gemm_v10.argtypes = lib8.cc_gemm_v8.argtypes
+    return lib, lib8, lib10
 
 
 def time_variant(fn, iters=30) -> float:
@@ -72,9 +82,9 @@ def time_variant(fn, iters=30) -> float:
 
 def main() -> None:
     prod = hotpath.require_gpu()
-    var, var8 = build()
+    var, var8, var10 = build()
     stream = torch.cuda.current_stream().cuda_stream
-    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v9':>7s} {'v8':>7s}  TF/s (best of 3 reps)")
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v9':>7s} {'v8':>7s} {'v10':>7s}  TF/s (best of 3 reps)")
     for label, M, N, K in SHAPES:
         torch.manual_seed(1)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
@@ -100,6 +110,11 @@ def main() -> None:
             rc = var8.cc_gemm_v8(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
             assert rc == 0
         calls[8] = v8call
+
+        def v10call():
+            rc = var10.cc_gemm_v10(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
+            assert rc == 0
+        calls[10] = v10call
         for v, fn in calls.items():
             if v != 0:
                 c.zero_()
