@@ -30,6 +30,11 @@ SHAPES = [
     ("fc2", 336 * 50, 768, 3072),
     ("square4k", 4096, 4096, 4096),
     ("square8k", 8192, 8192, 8192),
+    ("patch64", 1344 * 49, 768, 3072),
+    ("qkv64", 1344 * 50, 2304, 768),
+    ("out64", 1344 * 50, 768, 768),
+    ("fc1_64", 1344 * 50, 3072, 768),
+    ("fc2_64", 1344 * 50, 768, 3072),
 ]
 
 
