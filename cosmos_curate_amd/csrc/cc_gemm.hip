@@ -358,6 +358,216 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
 #undef BS32
 }
 
+// ---- 256x256 8-phase template kernel (guide "The 256² 8-phase
+// template", schedule derived + race-screened in tools/gemm_v10.hip,
+// measured profiles/r01_v10_b64.log) ----
+// 512 threads = 8 waves (2M x 4N), per-wave 128x64 = 8x4 frags; LDS
+// 128 KiB (1 WG/CU); raw s_barrier phases, counted vmcnt(4) at phases
+// 4/8 only, B fragments held in registers per K-tile.  Wins the
+// K>=1536 or N>=1536 shapes (patch/fc2 ~900, qkv/fc1 ~730-770 TF at
+// the batch-64 bench shapes vs ~700-830 for the 128² body) by halving
+// the per-launch A/B HBM re-read traffic, which bounds those shapes.
+constexpr int BM2 = 256, BN2 = 256;
+constexpr int WM2 = 128, WN2 = 64;
+constexpr int MFR2 = WM2 / FRAG;  // 8
+constexpr int NFR2 = WN2 / FRAG;  // 4
+
+// one wave stages its 16-row slice of a 128-row half-tile: 2 glds.
+__device__ __forceinline__ void stage_half(const __bf16* __restrict__ src,
+                                           long ld, long row0,
+                                           long row_limit, long k0,
+                                           __bf16* lds_rowbase, int lane) {
+  const int lrow8 = lane >> 3;
+  const int slot = lane & 7;
+  const int gk16 = slot ^ lrow8;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    long grow = row0 + j * 8 + lrow8;
+    grow = grow < 0 ? 0 : (grow >= row_limit ? row_limit - 1 : grow);
+    const __bf16* gptr = src + grow * ld + k0 + (long)gk16 * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gptr,
+        (__attribute__((address_space(3))) unsigned int*)(lds_rowbase +
+                                                          j * 8 * BK),
+        16, 0, 0);
+  }
+}
+
+template <int ACT, bool HAS_BIAS, bool HAS_RES>
+__global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    const __bf16* __restrict__ residual, long M, long N, long K,
+    int c_is_bf16, int nbx, int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM2 + BN2) * BK];  // 128 KiB
+#define A2T(b) (lds + (b) * (BM2 * BK))
+#define B2T(b) (lds + 2 * (BM2 * BK) + (b) * (BN2 * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  int orig = blockIdx.x;
+  if (do_remap) {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = orig & 7, lid = orig >> 3;
+    orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+  }
+  const long bm = (long)(orig / nbx) * BM2;
+  const long bn = (long)(orig % nbx) * BN2;
+
+  f32x4 acc[MFR2][NFR2] = {};
+  const long KT = K / BK;  // launcher guarantees even, >= 4
+  const long srow = wid * 16;
+
+#define STAGE_A2(t, h)                                                      \
+  stage_half(A, K, bm + (h) * 128 + srow, M, (t) * BK,                      \
+             A2T((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B2(t, h)                                                      \
+  stage_half(B, K, bn + (h) * 128 + srow, N, (t) * BK,                      \
+             B2T((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+  // prologue: A(0) h0,h1, B(0) h0,h1, B(1) h0,h1 (12 instrs); wait for
+  // tile 0's 8 (the 4 newest = B(1) stay in flight), publish.
+  STAGE_A2(0, 0);
+  STAGE_A2(0, 1);
+  STAGE_B2(0, 0);
+  STAGE_B2(0, 1);
+  STAGE_B2(1, 0);
+  STAGE_B2(1, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int arow_base = waveM * WM2 + (lane & 15);
+  const int brow_base = waveN * WN2 + (lane & 15);
+  bf16x8 bfragT[NFR2][2];  // held for the whole K-tile
+  bf16x8 afrag[2][2];      // one quadrant: 2 M-frags x 2 k-steps
+
+#define PHASE_MFMA2(q)                                                      \
+  do {                                                                      \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR2; n++) {                  \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+    __builtin_amdgcn_s_barrier();                                           \
+  } while (0)
+
+#define READ_A2(At, q)                                                      \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =             \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B2(Bt)                                                         \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR2; n++) bfragT[n][g] =         \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+// one K-tile = 4 phases; stages the next A halves in ph1-2 and the
+// tile-after-next B halves in ph3-4 (issue order proves vmcnt(4)):
+#define KTILE2(t)                                                           \
+  do {                                                                      \
+    const __bf16* At = A2T((t) & 1);                                        \
+    const __bf16* Bt = B2T((t) & 1);                                        \
+    READ_B2(Bt);                                                            \
+    READ_A2(At, 0);                                                         \
+    if ((t) + 1 < KT) STAGE_A2((t) + 1, 0);                                 \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA2(0);                                                         \
+    READ_A2(At, 1);                                                         \
+    if ((t) + 1 < KT) STAGE_A2((t) + 1, 1);                                 \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA2(1);                                                         \
+    READ_A2(At, 2);                                                         \
+    if ((t) + 2 < KT) STAGE_B2((t) + 2, 0);                                 \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA2(2);                                                         \
+    READ_A2(At, 3);                                                         \
+    if ((t) + 2 < KT) STAGE_B2((t) + 2, 1);                                 \
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                        \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA2(3);                                                         \
+  } while (0)
+
+  for (long it = 0; it < KT / 2; ++it) {
+    KTILE2(2 * it);
+    KTILE2(2 * it + 1);
+  }
+
+  // epilogue: same C/D map as the 128² body, MFR2 x NFR2 fragments
+  const long crow_base = bm + waveM * WM2 + 4 * (lane >> 4);
+  const long ccol_base = bn + waveN * WN2 + (lane & 15);
+  const bool interior = (bm + BM2 <= M) && (bn + BN2 <= N);
+  if (interior) {
+#pragma unroll
+    for (int m = 0; m < MFR2; m++) {
+#pragma unroll
+      for (int n = 0; n < NFR2; n++) {
+        const long col = ccol_base + n * FRAG;
+        float bval = 0.0f;
+        if constexpr (HAS_BIAS) bval = bias[col];
+        float rv[4];
+        if constexpr (HAS_RES) {
+#pragma unroll
+          for (int r = 0; r < 4; r++)
+            rv[r] = (float)residual[(crow_base + m * FRAG + r) * N + col];
+        }
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const long row = crow_base + m * FRAG + r;
+          float v = acc[m][n][r] + bval;
+          if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+          if constexpr (HAS_RES) v += rv[r];
+          if (c_is_bf16)
+            ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+          else
+            ((float*)C)[row * N + col] = v;
+        }
+      }
+    }
+    return;
+  }
+#pragma unroll
+  for (int m = 0; m < MFR2; m++) {
+#pragma unroll
+    for (int n = 0; n < NFR2; n++) {
+      const long col = ccol_base + n * FRAG;
+      if (col >= N) continue;
+      float bval = 0.0f;
+      if constexpr (HAS_BIAS) bval = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const long row = crow_base + m * FRAG + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r] + bval;
+        if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+        if constexpr (HAS_RES) v += (float)residual[row * N + col];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+  }
+#undef A2T
+#undef B2T
+#undef STAGE_A2
+#undef STAGE_B2
+#undef PHASE_MFMA2
+#undef READ_A2
+#undef READ_B2
+#undef KTILE2
+}
+
 }  // namespace
 
 extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
@@ -387,11 +597,30 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
     return e ? atoi(e) : -1;
   }();
   const bool wide = wide_env >= 0 ? wide_env != 0 : false;
+  // 256² 8-phase template for the shapes its halved A/B re-read traffic
+  // pays on (measured profiles/r01_v10_b64.log: wins every shape with
+  // K>=1536 or N>=1536; loses only N=768 && K=768 at large M).
+  // CC_GEMM_TILE=0|1 forces the 128²/256² body for A/B experiments.
+  static const int tile_env = [] {
+    const char* e = getenv("CC_GEMM_TILE");
+    return e ? atoi(e) : -1;
+  }();
+  const bool t256_ok = (K % 128 == 0) && (K >= 256) && M > BM2 / 2;
+  const bool t256 =
+      !wide && t256_ok &&
+      (tile_env >= 0 ? tile_env != 0 : (K >= 1536 || N >= 1536));
   const bool hb = bias != nullptr;
   const bool hr = residual != nullptr;
   // XCD remap only for outputs that spill L3 (measured negative on the
   // L2-resident ViT shapes, profiles/r01_opt3)
   const int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
+  if (t256) {
+    nbx = (int)((N + BN2 - 1) / BN2);
+    nby = (int)((M + BM2 - 1) / BM2);
+    nwg = nbx * nby;
+    grid = dim3(nwg);
+    block = dim3(512);
+  }
 #define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
   hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \
                      (const __bf16*)A, (const __bf16*)B, C, bias,             \
@@ -413,6 +642,8 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   } while (0)
   if (wide)
     CC_DISPATCH(k_gemm_bf16_w32);
+  else if (t256)
+    CC_DISPATCH(k_gemm_bf16_t256);
   else
     CC_DISPATCH(k_gemm_bf16);
 #undef CC_DISPATCH

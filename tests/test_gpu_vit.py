@@ -47,6 +47,8 @@ def cc_gemm(lib, a_bf16, b_bf16, bias=None, out_bf16=True):
         (8232, 768, 3072),  # patch-embed GEMM at batch 168 frames
         (400, 2304, 768),   # fused QKV
         (33, 128, 64),      # ragged M
+        (300, 1600, 256),   # 256-tile body (N trigger), ragged M+N tails
+        (2000, 768, 1536),  # 256-tile body (K trigger), tail M
     ],
 )
 def test_gemm_vs_torch(lib, M, N, K):
@@ -69,9 +71,9 @@ def test_gemm_bf16_out_and_no_bias(lib):
     torch.testing.assert_close(got, want, rtol=2e-2, atol=5e-2)
 
 
-def test_gemm_fused_gelu_and_residual(lib):
+@pytest.mark.parametrize(("M", "N", "K"), [(200, 256, 128), (513, 1600, 1536)])
+def test_gemm_fused_gelu_and_residual(lib, M, N, K):
     torch.manual_seed(3)
-    M, N, K = 200, 256, 128
     a = torch.randn(M, K).to(torch.bfloat16).cuda()
     b = torch.randn(N, K).to(torch.bfloat16).cuda()
     bias = torch.randn(N).float().cuda()
