@@ -1,0 +1,53 @@
+"""Determinism repro for the 256-tile GEMM at the batch-64 shapes."""
+import ctypes
+import os
+import pathlib
+import sys
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
+import torch  # noqa: E402
+
+from cosmos_curate_amd import hotpath  # noqa: E402
+from tools.gemm_bench2 import build  # noqa: E402
+
+lib = hotpath.require_gpu()
+_, _, var10 = build()
+
+
+def run_shape(M, N, K, fn, label):
+    torch.manual_seed(11)
+    a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
+    b = (torch.randn(N, K) * 0.3).to(torch.bfloat16).cuda()
+    bias = torch.randn(N).float().cuda()
+    out = torch.zeros((M, N), dtype=torch.bfloat16, device="cuda")
+    outs = []
+    for i in range(6):
+        out.zero_()
+        fn(a, b, bias, out, M, N, K)
+        torch.cuda.synchronize()
+        outs.append(out.clone())
+    # compare all pairs vs run 1 (skip run 0 = cold)
+    base = outs[1]
+    msgs = []
+    for i, o in enumerate(outs):
+        if not torch.equal(o, base):
+            d = (o.float() - base.float()).abs()
+            msgs.append(f"run{i} differs: max {d.max().item():.3f} n={(d > 0).sum().item()}")
+    print(f"{label} {M}x{N}x{K}: {'DETERMINISTIC(1..5)' if not msgs else '; '.join(msgs)}")
+
+
+def prod_call(a, b, bias, out, M, N, K):
+    hotpath.check(lib.cc_gemm_bf16_ex(
+        a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K,
+        bias.data_ptr(), 1, 0, None, torch.cuda.current_stream().cuda_stream))
+
+
+def v10_call(a, b, bias, out, M, N, K):
+    rc = var10.cc_gemm_v10(a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K,
+                           1, torch.cuda.current_stream().cuda_stream)
+    assert rc == 0
+
+
+for (M, N, K) in [(65856, 768, 3072), (67200, 2304, 768)]:
+    run_shape(M, N, K, prod_call, f"prod[{os.environ.get('CC_GEMM_TILE','-')}]")
+    run_shape(M, N, K, v10_call, "v10-noremap")
