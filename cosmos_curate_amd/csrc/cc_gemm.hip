@@ -434,8 +434,8 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
   STAGE_B2(0, 1);
   STAGE_B2(1, 0);
   STAGE_B2(1, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // full drain: counted
+  __builtin_amdgcn_s_barrier();                     // waits are unsound (above)
 
   const int arow_base = waveM * WM2 + (lane & 15);
   const int brow_base = waveN * WN2 + (lane & 15);
@@ -488,12 +488,19 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(1);                                                         \
     READ_A2(At, 2);                                                         \
+    /* full drain BEFORE issuing B(t+2): everything ph5 needs (A(t+1),   */ \
+    /* B(t+1)) has been issued and must land; a COUNTED wait is unsound  */ \
+    /* here because VMEM completions can retire out of order under load  */ \
+    /* (measured: vmcnt(4) at ph4 was nondeterministic at >=774-block    */ \
+    /* grids, profiles/r01_t256_det2.log); the drain sits one phase      */ \
+    /* before the consumer and ahead of the B(t+2) issues, so 2 half-    */ \
+    /* tiles still span the tile boundary.                               */ \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
     if ((t) + 2 < KT) STAGE_B2((t) + 2, 0);                                 \
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(2);                                                         \
     READ_A2(At, 3);                                                         \
     if ((t) + 2 < KT) STAGE_B2((t) + 2, 1);                                 \
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                        \
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(3);                                                         \
   } while (0)

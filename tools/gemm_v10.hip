@@ -11,13 +11,13 @@
 //   ph1: ds_read B(t) all 4 N-frags x 2k (8) + A quadrant0 (4);
 //        glds A(t+1) half0                      [A-buf((t+1)&1) free]
 //   ph2: ds_read A q1; glds A(t+1) half1
-//   ph3: ds_read A q2; glds B(t+2) half0        [B-buf(t&1) free after ph1]
-//   ph4: ds_read A q3; glds B(t+2) half1; s_waitcnt vmcnt(4)
-//        [needed by ph5: A(t+1) (issued ph1-2), B(t+1) (issued in the
-//         previous iteration's ph7-8 / prologue); the 4 newest VMEM
-//         instructions are B(t+2) = allowed outstanding]
-//   ph5-8: same with t+1, staging A(t+2) h0,h1 then B(t+3) h0,h1,
-//        vmcnt(4) at ph8.
+//   ph3: ds_read A q2; s_waitcnt vmcnt(0) (full drain: everything ph5
+//        needs is already issued; counted waits were measurably unsound
+//        here — VMEM completions retire out of order under load, see
+//        profiles/r01_t256_det2.log); then glds B(t+2) half0
+//        [B-buf(t&1) free after ph1]
+//   ph4: ds_read A q3; glds B(t+2) half1  (B(t+2) spans the barrier)
+//   ph5-8: same with t+1, staging A(t+2) h0,h1 then B(t+3) h0,h1.
 // Each phase: [reads; glds] -> s_barrier -> lgkmcnt(0) -> setprio(1)
 // -> 16 MFMA -> setprio(0) -> s_barrier (raw barriers: no vmcnt(0)
 // glds drain, the template's whole point).
@@ -108,7 +108,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_v10(
   STAGE_B(0, 1);
   STAGE_B(1, 0);
   STAGE_B(1, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
   const int arow_base = waveM * WM + (lane & 15);
@@ -167,13 +167,13 @@ __global__ __launch_bounds__(512, 1) void k_gemm_v10(
       PHASE_MFMA(1);
       // ph3
       READ_A(At, 2);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // see prod note
       if (t + 2 < KT) STAGE_B(t + 2, 0);
       __builtin_amdgcn_s_barrier();
       PHASE_MFMA(2);
       // ph4
       READ_A(At, 3);
       if (t + 2 < KT) STAGE_B(t + 2, 1);
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       PHASE_MFMA(3);
     }
@@ -194,13 +194,13 @@ __global__ __launch_bounds__(512, 1) void k_gemm_v10(
       PHASE_MFMA(1);
       // ph7
       READ_A(At, 2);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       if (u + 2 < KT) STAGE_B(u + 2, 0);
       __builtin_amdgcn_s_barrier();
       PHASE_MFMA(2);
       // ph8
       READ_A(At, 3);
       if (u + 2 < KT) STAGE_B(u + 2, 1);
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       PHASE_MFMA(3);
     }
