@@ -620,13 +620,17 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   const bool hr = residual != nullptr;
   // XCD remap only for outputs that spill L3 (measured negative on the
   // L2-resident ViT shapes, profiles/r01_opt3)
-  const int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
+  int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
   if (t256) {
     nbx = (int)((N + BN2 - 1) / BN2);
     nby = (int)((M + BM2 - 1) / BM2);
     nwg = nbx * nby;
     grid = dim3(nwg);
     block = dim3(512);
+    // for the 256² body, row-band L2 sharing via the remap pays only on
+    // skinny-N grids (qkv64 +15%, fc1_64 +12%); on wide grids it costs
+    // (square8k 962 vs 1168 TF) — profiles/r01_t256_fixed.log
+    if (nbx > 12) remap = 0;
   }
 #define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
   hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \
