@@ -455,8 +455,14 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
       }                                                                     \
     }                                                                       \
     __builtin_amdgcn_s_setprio(0);                                          \
-    __builtin_amdgcn_s_barrier();                                           \
   } while (0)
+  /* NOTE: no trailing barrier per phase (v11, profiles/r01_v11_ab.log,
+     +5-8%): the B-buffer write-after-read hazard needs only 2-phase
+     separation, which the per-phase leading barrier provides — wave W
+     drains its reads (lgkm0) before it reaches the NEXT phase's
+     barrier, and any other wave's conflicting glds issues only after
+     passing that barrier, two phases later.  Race-screened +
+     determinism-checked at 774-3156-block grids. */
 
 #define READ_A2(At, q)                                                      \
   _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
