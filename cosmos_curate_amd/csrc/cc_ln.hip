@@ -4,8 +4,17 @@
 // calls/step at ~2 TB/s effective, ~10% of the bench step).  Stats in
 // f32 (torch layer_norm opmath semantics), bf16 in/out, f32 affine
 // params.  One wave per row: H/64 elements per lane, two shfl_xor
-// reduction trees (sum, sumsq), coalesced bf16x2 loads/stores.
-// H must be a multiple of 128 (768/1024/3072/4096 all comply).
+// reduction trees (sum, sumsq), bf16x4 (8 B) loads/stores (guide §5
+// rule 2: scalar/narrow bf16 access leaves >2x bandwidth on the table;
+// measured 4.7 TB/s at bf16x2).
+// H must be a multiple of 256 (768/1024/3072/4096 all comply).
+//
+// Also here: cc_embed_assemble_ln — the ViT embedding assembly
+// ([CLS; patch tokens] + position embedding, then pre-LN) fused into
+// one pass.  Replaces a torch cat + f32 add + bf16 cast + layer_norm
+// chain (4 kernel launches, ~300 us/step at the bench shape) with one
+// bandwidth-bound kernel (reference models/clip.py uses HF
+// CLIPVisionEmbeddings: cat + pos_embed add, then pre_layrnorm).
 
 #include <hip/hip_runtime.h>
 
@@ -17,6 +26,8 @@
 namespace {
 
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll
@@ -24,7 +35,7 @@ __device__ __forceinline__ float wave_sum(float v) {
   return v;
 }
 
-template <int CHUNK>  // elements per lane (H / 64)
+template <int CHUNK>  // elements per lane (H / 64), multiple of 4
 __global__ void k_layernorm_bf16(const __bf16* __restrict__ x,
                                  const float* __restrict__ w,
                                  const float* __restrict__ b,
@@ -40,13 +51,15 @@ __global__ void k_layernorm_bf16(const __bf16* __restrict__ x,
   float vals[CHUNK];
   float s = 0.0f, ss = 0.0f;
 #pragma unroll
-  for (int c = 0; c < CHUNK; c += 2) {
-    // coalesced: lane i reads elements (c*64 + 2i, +1)
-    bf16x2 p = *(const bf16x2*)(xr + c * 64 + 2 * lane);
-    vals[c] = (float)p.x;
-    vals[c + 1] = (float)p.y;
-    s += vals[c] + vals[c + 1];
-    ss += vals[c] * vals[c] + vals[c + 1] * vals[c + 1];
+  for (int c = 0; c < CHUNK; c += 4) {
+    // coalesced: lane i reads elements (c*64 + 4i .. +3)
+    bf16x4 p = *(const bf16x4*)(xr + c * 64 + 4 * lane);
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      vals[c + j] = (float)p[j];
+      s += vals[c + j];
+      ss += vals[c + j] * vals[c + j];
+    }
   }
   s = wave_sum(s);
   ss = wave_sum(ss);
@@ -55,14 +68,72 @@ __global__ void k_layernorm_bf16(const __bf16* __restrict__ x,
   const float var = ss * inv_n - mean * mean;
   const float rstd = rsqrtf(var + eps);
 #pragma unroll
-  for (int c = 0; c < CHUNK; c += 2) {
-    const int i0 = c * 64 + 2 * lane;
-    float w0 = w[i0], w1 = w[i0 + 1];
-    float b0 = b[i0], b1 = b[i0 + 1];
-    bf16x2 o;
-    o.x = (__bf16)((vals[c] - mean) * rstd * w0 + b0);
-    o.y = (__bf16)((vals[c + 1] - mean) * rstd * w1 + b1);
-    *(bf16x2*)(yr + i0) = o;
+  for (int c = 0; c < CHUNK; c += 4) {
+    const int i0 = c * 64 + 4 * lane;
+    f32x4 wv = *(const f32x4*)(w + i0);
+    f32x4 bv = *(const f32x4*)(b + i0);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; j++)
+      o[j] = (__bf16)((vals[c + j] - mean) * rstd * wv[j] + bv[j]);
+    *(bf16x4*)(yr + i0) = o;
+  }
+}
+
+// out[row] = LN( src(row) + pos[t] ) where row = f*tokens + t,
+// src = cls (t==0) or tok[f*(tokens-1) + t - 1]; add in f32 like the
+// torch chain it replaces (h.float() + pos -> bf16 -> layer_norm).
+template <int CHUNK>
+__global__ void k_embed_assemble_ln(const __bf16* __restrict__ tok,
+                                    const float* __restrict__ cls,
+                                    const float* __restrict__ pos,
+                                    const float* __restrict__ w,
+                                    const float* __restrict__ b,
+                                    __bf16* __restrict__ y, long nrows,
+                                    int tokens, int H, float eps) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long row = (long)blockIdx.x * 4 + wave;
+  if (row >= nrows) return;
+  const long f = row / tokens;
+  const int t = (int)(row - f * tokens);
+  const __bf16* tr = (t == 0) ? nullptr : tok + (f * (tokens - 1) + t - 1) * H;
+  const float* pr = pos + (long)t * H;
+  __bf16* yr = y + row * H;
+
+  float vals[CHUNK];
+  float s = 0.0f, ss = 0.0f;
+#pragma unroll
+  for (int c = 0; c < CHUNK; c += 4) {
+    const int i0 = c * 64 + 4 * lane;
+    f32x4 pv = *(const f32x4*)(pr + i0);
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      // cls follows the torch path: f32 param cast to bf16 first
+      float base = tr ? (float)tr[i0 + j] : (float)(__bf16)cls[i0 + j];
+      // match torch: (f32 add) rounded to bf16, stats on the bf16 value
+      float v = (float)(__bf16)(base + pv[j]);
+      vals[c + j] = v;
+      s += v;
+      ss += v * v;
+    }
+  }
+  s = wave_sum(s);
+  ss = wave_sum(ss);
+  const float inv_n = 1.0f / (float)H;
+  const float mean = s * inv_n;
+  const float var = ss * inv_n - mean * mean;
+  const float rstd = rsqrtf(var + eps);
+#pragma unroll
+  for (int c = 0; c < CHUNK; c += 4) {
+    const int i0 = c * 64 + 4 * lane;
+    f32x4 wv = *(const f32x4*)(w + i0);
+    f32x4 bv = *(const f32x4*)(b + i0);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; j++)
+      o[j] = (__bf16)((vals[c + j] - mean) * rstd * wv[j] + bv[j]);
+    *(bf16x4*)(yr + i0) = o;
   }
 }
 
@@ -73,8 +144,8 @@ extern "C" int cc_layernorm_bf16(const void* x, const void* w, const void* b,
                                  uint64_t stream) {
   if (!x || !w || !b || !y || M <= 0 || H <= 0)
     return cc::set_error(CC_ERR_INVALID, "bad layernorm args");
-  if (H % 128 != 0 || H > 64 * 64)
-    return cc::set_error(CC_ERR_UNSUPPORTED, "H must be k*128, <=4096 (got %lld)",
+  if (H % 256 != 0 || H > 64 * 64)
+    return cc::set_error(CC_ERR_UNSUPPORTED, "H must be k*256, <=4096 (got %lld)",
                          (long long)H);
   dim3 block(256), grid((M + 3) / 4);
   hipEvent_t ev0, ev1;
@@ -87,7 +158,7 @@ extern "C" int cc_layernorm_bf16(const void* x, const void* w, const void* b,
                        (const float*)w, (const float*)b, (__bf16*)y,       \
                        (long)M, (int)H, eps);                              \
     break;
-    CASE(2) CASE(4) CASE(6) CASE(8) CASE(12) CASE(16) CASE(32) CASE(48) CASE(64)
+    CASE(4) CASE(8) CASE(12) CASE(16) CASE(32) CASE(48) CASE(64)
 #undef CASE
     default:
       return cc::set_error(CC_ERR_UNSUPPORTED, "unsupported H %lld", (long long)H);
@@ -96,5 +167,42 @@ extern "C" int cc_layernorm_bf16(const void* x, const void* w, const void* b,
   if (timed) cc::timed_end("layernorm_bf16", stream, ev0, ev1);
   if (e != hipSuccess)
     return cc::set_error(CC_ERR_HIP, "layernorm launch: %s", hipGetErrorString(e));
+  return CC_OK;
+}
+
+extern "C" int cc_embed_assemble_ln(const void* tok, const void* cls,
+                                    const void* pos, const void* w,
+                                    const void* b, void* y, int64_t n_frames,
+                                    int64_t tokens, int64_t H, float eps,
+                                    uint64_t stream) {
+  if (!tok || !cls || !pos || !w || !b || !y || n_frames <= 0 || tokens <= 1 ||
+      H <= 0)
+    return cc::set_error(CC_ERR_INVALID, "bad embed_assemble args");
+  if (H % 256 != 0 || H > 64 * 64)
+    return cc::set_error(CC_ERR_UNSUPPORTED, "H must be k*256, <=4096 (got %lld)",
+                         (long long)H);
+  const long nrows = (long)n_frames * tokens;
+  dim3 block(256), grid((nrows + 3) / 4);
+  hipEvent_t ev0, ev1;
+  bool timed = cc::timed_begin(stream, &ev0, &ev1);
+  switch (H / 64) {
+#define CASE(C)                                                            \
+  case C:                                                                  \
+    hipLaunchKernelGGL(k_embed_assemble_ln<C>, grid, block, 0,             \
+                       (hipStream_t)stream, (const __bf16*)tok,            \
+                       (const float*)cls, (const float*)pos,               \
+                       (const float*)w, (const float*)b, (__bf16*)y,       \
+                       nrows, (int)tokens, (int)H, eps);                   \
+    break;
+    CASE(4) CASE(8) CASE(12) CASE(16) CASE(32) CASE(48) CASE(64)
+#undef CASE
+    default:
+      return cc::set_error(CC_ERR_UNSUPPORTED, "unsupported H %lld", (long long)H);
+  }
+  hipError_t e = hipGetLastError();
+  if (timed) cc::timed_end("embed_assemble_ln", stream, ev0, ev1);
+  if (e != hipSuccess)
+    return cc::set_error(CC_ERR_HIP, "embed_assemble launch: %s",
+                         hipGetErrorString(e));
   return CC_OK;
 }

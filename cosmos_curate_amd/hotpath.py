@@ -77,6 +77,9 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.cc_layernorm_bf16.argtypes = [
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64,
         c.c_float, c.c_uint64]
+    lib.cc_embed_assemble_ln.argtypes = [
+        c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p,
+        c.c_void_p, c.c_int64, c.c_int64, c.c_int64, c.c_float, c.c_uint64]
     lib.cc_pairwise_max_earlier.argtypes = [
         c.c_void_p, c.c_int64, c.c_int64, c.c_void_p, c.c_void_p, c.c_uint64]
 

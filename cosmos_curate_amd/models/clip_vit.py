@@ -171,13 +171,27 @@ class ClipVisionTowerAMD(torch.nn.Module):
         )
         if self.patch_k != k0:
             patches = torch.nn.functional.pad(patches, (0, self.patch_k - k0))
-        tok = self._linear(patches, self.w_patch, None).reshape(n, g * g, cfg.hidden)
-        cls = self.cls_emb.to(tok.dtype).expand(n, 1, cfg.hidden)
-        h = torch.cat([cls, tok], dim=1)  # (n, tokens, hidden)
-        h = (h.float() + self.pos_emb.unsqueeze(0)).to(torch.bfloat16)
-        h = self._ln(h, self.pre_ln_w, self.pre_ln_b)
-
-        seq = h.shape[1]
+        tok_flat = self._linear(patches, self.w_patch, None)  # (n*g*g, hidden)
+        seq = g * g + 1
+        if tok_flat.is_cuda:
+            # fused [CLS; tok] + pos-embed + pre-LN (csrc/cc_ln.hip) — one
+            # bandwidth pass instead of the cat/f32-add/cast/LN chain
+            lib = hotpath.require_gpu()
+            h = torch.empty((n * seq, cfg.hidden), dtype=torch.bfloat16,
+                            device=tok_flat.device)
+            stream = torch.cuda.current_stream(tok_flat.device).cuda_stream
+            hotpath.check(lib.cc_embed_assemble_ln(
+                tok_flat.data_ptr(), self.cls_emb.data_ptr(),
+                self.pos_emb.data_ptr(), self.pre_ln_w.data_ptr(),
+                self.pre_ln_b.data_ptr(), h.data_ptr(), n, seq, cfg.hidden,
+                ctypes.c_float(1e-5), stream))
+            h = h.reshape(n, seq, cfg.hidden)
+        else:
+            tok = tok_flat.reshape(n, g * g, cfg.hidden)
+            cls = self.cls_emb.to(tok.dtype).expand(n, 1, cfg.hidden)
+            h = torch.cat([cls, tok], dim=1)  # (n, tokens, hidden)
+            h = (h.float() + self.pos_emb.unsqueeze(0)).to(torch.bfloat16)
+            h = self._ln(h, self.pre_ln_w, self.pre_ln_b)
         hd = cfg.hidden // self.heads
         for i in range(self.layers):
             res = h

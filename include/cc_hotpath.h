@@ -164,9 +164,20 @@ int cc_attn_small(const void* qkv, void* out, int64_t n_frames, int seq,
                   int heads, int hidden, float scale, uint64_t stream);
 
 /* ---- fused bf16 LayerNorm (replaces torch layer_norm in the ViT
- * forward; f32 stats/affine, H multiple of 128). */
+ * forward; f32 stats/affine, H multiple of 256). */
 int cc_layernorm_bf16(const void* x, const void* w, const void* b, void* y,
                       int64_t M, int64_t H, float eps, uint64_t stream);
+
+/* ---- fused ViT embedding assembly: out[f*tokens + t] =
+ * LayerNorm( (t==0 ? cls : tok[f*(tokens-1)+t-1]) + pos[t] ).
+ * Replaces the HF CLIPVisionEmbeddings cat + position-embedding add +
+ * pre_layrnorm chain (reference models/clip.py get_image_features
+ * entry).  tok bf16 [n_frames*(tokens-1), H]; cls/pos/w/b f32; out bf16
+ * [n_frames*tokens, H]; H multiple of 256. */
+int cc_embed_assemble_ln(const void* tok, const void* cls, const void* pos,
+                         const void* w, const void* b, void* y,
+                         int64_t n_frames, int64_t tokens, int64_t H,
+                         float eps, uint64_t stream);
 
 /* ---- semantic dedup -------------------------------------------------
  * Strict-upper-triangular max-cosine scan (SemDedupActor.dedup,

@@ -252,3 +252,30 @@ def test_iv2_preprocess_matches_oracle(lib):
         )
     with pytest.raises(ValueError):
         temporal_subsample(5, 8)
+
+
+def test_embed_assemble_ln_vs_torch(lib):
+    """cc_embed_assemble_ln == cat + f32 pos-add + bf16 cast + LN chain."""
+    import ctypes
+
+    torch.manual_seed(17)
+    n, tokens, H = 7, 50, 768
+    tok = torch.randn(n * (tokens - 1), H).to(torch.bfloat16).cuda()
+    cls = torch.randn(H).float().cuda()
+    pos = torch.randn(tokens, H).float().cuda()
+    w = torch.randn(H).float().cuda()
+    b = torch.randn(H).float().cuda()
+    out = torch.empty(n * tokens, H, dtype=torch.bfloat16, device="cuda")
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(lib.cc_embed_assemble_ln(
+        tok.data_ptr(), cls.data_ptr(), pos.data_ptr(), w.data_ptr(),
+        b.data_ptr(), out.data_ptr(), n, tokens, H, ctypes.c_float(1e-5),
+        stream))
+    torch.cuda.synchronize()
+
+    h = torch.cat([cls.to(torch.bfloat16).expand(n, 1, H),
+                   tok.reshape(n, tokens - 1, H)], dim=1)
+    h = (h.float() + pos.unsqueeze(0)).to(torch.bfloat16)
+    want = torch.nn.functional.layer_norm(
+        h.float(), (H,), w, b, eps=1e-5).reshape(n * tokens, H)
+    torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
