@@ -137,6 +137,32 @@ __global__ void k_embed_assemble_ln(const __bf16* __restrict__ tok,
   }
 }
 
+// narrow rows (H=128: 2 elements/lane) keep the bf16x2 form
+__global__ void k_layernorm_bf16_h128(const __bf16* __restrict__ x,
+                                      const float* __restrict__ w,
+                                      const float* __restrict__ b,
+                                      __bf16* __restrict__ y, long M,
+                                      float eps) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long row = (long)blockIdx.x * 4 + wave;
+  if (row >= M) return;
+  const __bf16* xr = x + row * 128;
+  __bf16* yr = y + row * 128;
+  bf16x2 p = *(const bf16x2*)(xr + 2 * lane);
+  float v0 = (float)p.x, v1 = (float)p.y;
+  float s = wave_sum(v0 + v1);
+  float ss = wave_sum(v0 * v0 + v1 * v1);
+  const float mean = s / 128.0f;
+  const float var = ss / 128.0f - mean * mean;
+  const float rstd = rsqrtf(var + eps);
+  const int i0 = 2 * lane;
+  bf16x2 o;
+  o.x = (__bf16)((v0 - mean) * rstd * w[i0] + b[i0]);
+  o.y = (__bf16)((v1 - mean) * rstd * w[i0 + 1] + b[i0 + 1]);
+  *(bf16x2*)(yr + i0) = o;
+}
+
 }  // namespace
 
 extern "C" int cc_layernorm_bf16(const void* x, const void* w, const void* b,
@@ -144,13 +170,20 @@ extern "C" int cc_layernorm_bf16(const void* x, const void* w, const void* b,
                                  uint64_t stream) {
   if (!x || !w || !b || !y || M <= 0 || H <= 0)
     return cc::set_error(CC_ERR_INVALID, "bad layernorm args");
-  if (H % 256 != 0 || H > 64 * 64)
-    return cc::set_error(CC_ERR_UNSUPPORTED, "H must be k*256, <=4096 (got %lld)",
+  if ((H % 256 != 0 && H != 128) || H > 64 * 64)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "H must be 128 or k*256, <=4096 (got %lld)",
                          (long long)H);
   dim3 block(256), grid((M + 3) / 4);
   hipEvent_t ev0, ev1;
   bool timed = cc::timed_begin(stream, &ev0, &ev1);
   switch (H / 64) {
+    case 2:
+      hipLaunchKernelGGL(k_layernorm_bf16_h128, grid, block, 0,
+                         (hipStream_t)stream, (const __bf16*)x,
+                         (const float*)w, (const float*)b, (__bf16*)y,
+                         (long)M, eps);
+      break;
 #define CASE(C)                                                            \
   case C:                                                                  \
     hipLaunchKernelGGL(k_layernorm_bf16<C>, grid, block, 0,                \
