@@ -74,6 +74,7 @@ def build() -> ctypes.CDLL:
     lib10 = ctypes.CDLL(str(SO10))
     lib10.cc_gemm_v10.argtypes = lib8.cc_gemm_v8.argtypes
     lib10.cc_gemm_v11.argtypes = lib8.cc_gemm_v8.argtypes
+    lib10.cc_gemm_v12.argtypes = lib8.cc_gemm_v8.argtypes
     return lib, lib8, lib10
 
 
@@ -90,7 +91,7 @@ def main() -> None:
     prod = hotpath.require_gpu()
     var, var8, var10 = build()
     stream = torch.cuda.current_stream().cuda_stream
-    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v9':>7s} {'v8':>7s} {'v10':>7s} {'v11':>7s}  TF/s (best of 3 reps)")
+    print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} | {'prod':>7s} {'v2':>7s} {'v6':>7s} {'v9':>7s} {'v8':>7s} {'v10':>7s} {'v11':>7s} {'v12':>7s}  TF/s (best of 3 reps)")
     for label, M, N, K in SHAPES:
         torch.manual_seed(1)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
@@ -126,6 +127,11 @@ def main() -> None:
             rc = var10.cc_gemm_v11(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
             assert rc == 0
         calls[11] = v11call
+
+        def v12call():
+            rc = var10.cc_gemm_v12(a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, stream)
+            assert rc == 0
+        calls[12] = v12call
         for v, fn in calls.items():
             if v != 0:
                 c.zero_()

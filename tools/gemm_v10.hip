@@ -254,6 +254,155 @@ extern "C" int cc_gemm_v10(const void* A, const void* B, void* C, int64_t M,
   return CC_OK;
 }
 
+
+// ---- V12: v11 + A-quadrant reads software-pipelined into the MFMA
+// region.  afrag is double-buffered (+16 VGPR): the reads for quadrant
+// q+1 are issued textually after MFMA(q) whose operands live in the
+// OTHER afrag set, so the scheduler interleaves them under the MFMAs
+// and the next phase's lgkmcnt(0) is nearly free.  B reads and glds
+// keep the v11 placement.  Sync structure: same barriers/drains as v11.
+namespace v12 {
+constexpr int WM = 128, WN = 64, FRAG = 16;
+constexpr int MFR = WM / FRAG, NFR = WN / FRAG;
+
+__global__ __launch_bounds__(512, 1) void k_gemm_v12(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];
+#define A12(b) (lds + (b) * (BM * BK))
+#define B12(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long bm = (long)blockIdx.y * BM;
+  const long bn = (long)blockIdx.x * BN;
+
+  f32x4 acc[MFR][NFR] = {};
+  const long KT = K / BK;
+  const long srow = wid * 16;
+
+#define STG_A(t, h)                                                         \
+  stage_half(A, K, bm + (h) * 128 + srow, M, (t) * BK,                      \
+             A12((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STG_B(t, h)                                                         \
+  stage_half(B, K, bn + (h) * 128 + srow, N, (t) * BK,                      \
+             B12((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+  STG_A(0, 0);
+  STG_A(0, 1);
+  STG_B(0, 0);
+  STG_B(0, 1);
+  STG_B(1, 0);
+  STG_B(1, 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2][2];  // [set][m][g]
+
+#define RD_A(At, q, set)                                                    \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[set][m][g] =        \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+#define RD_B(Bt)                                                            \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+#define MM(q, set)                                                          \
+  _Pragma("unroll") for (int g = 0; g < 2; g++)                             \
+  _Pragma("unroll") for (int m = 0; m < 2; m++)                             \
+  _Pragma("unroll") for (int n = 0; n < NFR; n++)                           \
+      acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(        \
+          afrag[set][m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);
+
+  // one K-tile, phases 0..3.  Phase p: BARRIER; lgkm0; MFMA(q_p) with
+  // next reads + this phase's glds interleaved under it.
+#define TILE12(t)                                                           \
+  do {                                                                      \
+    const __bf16* At = A12((t) & 1);                                        \
+    const __bf16* Atn = A12(((t) + 1) & 1);                                 \
+    /* ph0 (q0 reads were issued in the previous tile's ph3 / preloop) */   \
+    __builtin_amdgcn_s_barrier();                                           \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MM(0, 0);                                                               \
+    RD_A(At, 1, 1);                                                         \
+    if ((t) + 1 < KT) STG_A((t) + 1, 0);                                    \
+    __builtin_amdgcn_s_setprio(0);                                          \
+    /* ph1 */                                                               \
+    __builtin_amdgcn_s_barrier();                                           \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MM(1, 1);                                                               \
+    RD_A(At, 2, 0);                                                         \
+    if ((t) + 1 < KT) STG_A((t) + 1, 1);                                    \
+    __builtin_amdgcn_s_setprio(0);                                          \
+    /* ph2: drain BEFORE B(t+2) issues (everything ph0(t+1) needs is    */  \
+    /* in flight already; counted waits unsound — see v10 header)       */  \
+    __builtin_amdgcn_s_barrier();                                           \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MM(2, 0);                                                               \
+    RD_A(At, 3, 1);                                                         \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    if ((t) + 2 < KT) STG_B((t) + 2, 0);                                    \
+    __builtin_amdgcn_s_setprio(0);                                          \
+    /* ph3: B(t+1)/A(t+1,q0) reads issue after the last MFMA consuming  */  \
+    /* bfragT/afrag — register deps order them; glds B(t+2)h1 follows   */  \
+    __builtin_amdgcn_s_barrier();                                           \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MM(3, 1);                                                               \
+    if ((t) + 1 < KT) {                                                     \
+      RD_B(B12(((t) + 1) & 1));                                             \
+      RD_A(Atn, 0, 0);                                                      \
+    }                                                                       \
+    if ((t) + 2 < KT) STG_B((t) + 2, 1);                                    \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+  RD_B(B12(0));
+  RD_A(A12(0), 0, 0);
+  for (long t = 0; t < KT; ++t) TILE12(t);
+
+  const long crow_base = bm + waveM * WM + 4 * (lane >> 4);
+  const long ccol_base = bn + waveN * WN + (lane & 15);
+#pragma unroll
+  for (int m = 0; m < MFR; m++) {
+#pragma unroll
+    for (int n = 0; n < NFR; n++) {
+      const long col = ccol_base + n * FRAG;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; r++) {
+        const long row = crow_base + m * FRAG + r;
+        if (row >= M) continue;
+        float v = acc[m][n][r];
+        if (c_is_bf16)
+          ((unsigned short*)C)[row * N + col] = bf16_rne(v);
+        else
+          ((float*)C)[row * N + col] = v;
+      }
+    }
+  }
+#undef A12
+#undef B12
+#undef STG_A
+#undef STG_B
+#undef RD_A
+#undef RD_B
+#undef MM
+#undef TILE12
+}
+}  // namespace v12
+
 // v11 = v10 with ONE barrier per phase (the trailing barrier dropped:
 // the B-buffer write-after-read hazard only needs 2-phase separation,
 // which the per-phase leading barrier already provides — see the
@@ -267,6 +416,21 @@ extern "C" int cc_gemm_v11(const void* A, const void* B, void* C, int64_t M,
   dim3 block(512);
   dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
   hipLaunchKernelGGL((k_gemm_v10<false>), grid, block, 0, (hipStream_t)stream,
+                     (const __bf16*)A, (const __bf16*)B, C, (long)M, (long)N,
+                     (long)K, c_dtype == 1);
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess) return cc::set_error(CC_ERR_HIP, "%s", hipGetErrorString(e));
+  return CC_OK;
+}
+
+extern "C" int cc_gemm_v12(const void* A, const void* B, void* C, int64_t M,
+                           int64_t N, int64_t K, int c_dtype,
+                           uint64_t stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4)
+    return cc::set_error(CC_ERR_UNSUPPORTED, "K must be multiple of 128, >=256");
+  dim3 block(512);
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+  hipLaunchKernelGGL(v12::k_gemm_v12, grid, block, 0, (hipStream_t)stream,
                      (const __bf16*)A, (const __bf16*)B, C, (long)M, (long)N,
                      (long)K, c_dtype == 1);
   hipError_t e = hipGetLastError();

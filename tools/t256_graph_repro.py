@@ -54,7 +54,14 @@ def v11_call(a, b, bias, out, M, N, K):
     assert rc == 0
 
 
+def v12_call(a, b, bias, out, M, N, K):
+    rc = var10.cc_gemm_v12(a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K,
+                           1, torch.cuda.current_stream().cuda_stream)
+    assert rc == 0
+
+
 for (M, N, K) in [(65856, 768, 3072), (67200, 2304, 768), (8192, 8192, 8192)]:
     run_shape(M, N, K, prod_call, f"prod[{os.environ.get('CC_GEMM_TILE','-')}]")
     run_shape(M, N, K, v10_call, "v10-noremap")
     run_shape(M, N, K, v11_call, "v11-1barrier")
+    run_shape(M, N, K, v12_call, "v12-pipelined")
