@@ -200,3 +200,17 @@ class ClipFrameExtractionStage(CuratorStage):
                 name, stats = self._timer.log_stats()
                 task.stage_perf[name] = stats
         return tasks
+
+def extract_frames(
+    data: bytes,
+    *,
+    sample_rate_fps: float,
+    target_res: tuple[int, int] = (-1, -1),
+    to_host: bool = False,
+) -> "torch.Tensor | np.ndarray":
+    """One-off frame extraction (reference decoder_utils.extract_frames
+    shape, used by the embedding stage's doubling-fps re-extraction,
+    internvideo2_stages.py:157-175)."""
+    stage = ClipFrameExtractionStage(target_res=target_res, to_host=to_host)
+    frames = stage._extract_clip_frames(data, float(sample_rate_fps))
+    return frames.cpu().numpy() if to_host else frames

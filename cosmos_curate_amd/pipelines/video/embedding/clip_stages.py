@@ -34,6 +34,9 @@ from cosmos_curate_amd.core.utils.lazy_data import LazyData
 from cosmos_curate_amd.core.utils.performance_utils import StageTimer
 from cosmos_curate_amd.core.utils.roctx import annotate
 from cosmos_curate_amd.models.clip import CLIPImageEmbeddings
+from cosmos_curate_amd.pipelines.video.clipping.clip_frame_extraction_stages import (
+    extract_frames,
+)
 from cosmos_curate_amd.pipelines.video.utils.data_model import Clip, SplitPipeTask
 from cosmos_curate_amd.pipelines.video.utils.decoder_utils import (
     FrameExtractionPolicy,
@@ -49,12 +52,14 @@ class ClipFrameCreationStage(CuratorStage):
         target_fps: float = 2.0,
         *,
         max_frames: int | None = None,
+        min_frames: int | None = None,
         verbose: bool = False,
         log_stats: bool = False,
     ) -> None:
         self._timer = StageTimer(self)
         self._target_fps = target_fps
         self._max_frames = max_frames
+        self._min_frames = min_frames
         self._verbose = verbose
         self._log_stats = log_stats
 
@@ -73,6 +78,22 @@ class ClipFrameCreationStage(CuratorStage):
             clip.errors["clip_embedding_frames"] = f"missing signature {sig}"
             return
         frames = frames_map[sig]
+        # frame-count guarantee: re-extract at doubling fps (<=20) until the
+        # model's required count is reached (internvideo2_stages.py:137-175;
+        # on running out of fps headroom the reference logs and keeps what it
+        # has — no per-clip error)
+        if self._min_frames is not None and len(frames) < self._min_frames:
+            data = clip.encoded_data.resolve()
+            regen_fps = self._target_fps
+            while data is not None and len(frames) < self._min_frames:
+                regen_fps *= 2
+                if regen_fps > 20:
+                    break
+                raw = bytes(data) if not isinstance(data, bytes) else data
+                frames = extract_frames(
+                    raw, sample_rate_fps=regen_fps,
+                    to_host=isinstance(frames, np.ndarray),
+                )
         if self._max_frames is not None and len(frames) > self._max_frames:
             step = len(frames) // self._max_frames  # IV2 step rule (:400-401)
             frames = frames[::step][: self._max_frames]

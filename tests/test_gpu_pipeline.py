@@ -171,3 +171,35 @@ def test_extraction_full_1080p_clip_bitexact():
         rgb = ocolor.nv12_to_rgb(ys[0], uvs[0].reshape(h // 2, w // 2, 2))
         want = ocolor.resize_bilinear_u8(rgb, 224, 224)
         np.testing.assert_array_equal(frames[j].cpu().numpy(), want)
+
+
+def test_frame_creation_doubling_fps_guarantee():
+    """Frame-count guarantee (internvideo2_stages.py:137-175): a short clip
+    with too few frames at target_fps is re-extracted at doubling fps
+    (<=20) until min_frames is reached."""
+    import uuid as uuid_mod
+
+    h, w, fps, secs = 64, 96, 30, 2  # 2 s clip: 2 fps -> 5 frames only
+    raw = raw_backend.make_synthetic_clip(fps * secs, h, w, fps, seed=3)
+    v = Video(
+        input_video=pathlib.Path("/synthetic/short.nv12"),
+        metadata=VideoMetadata(size=1, height=h, width=w, framerate=float(fps),
+                               num_frames=fps * secs, duration=float(secs),
+                               video_codec="raw"),
+    )
+    v.clips.append(
+        Clip(uuid=uuid_mod.uuid4(), source_video="s", span=(0.0, float(secs)),
+             encoded_data=np.frombuffer(raw, dtype=np.uint8))
+    )
+    task = SplitPipeTask(videos=[v])
+    out = run_pipeline(
+        [task],
+        [ClipFrameExtractionStage(target_fps=[2], target_res=(64, 96)),
+         ClipFrameCreationStage(target_fps=2.0, min_frames=8)],
+        runner=SequentialRunner(),
+    )
+    clip = out[0].video.clips[0]
+    assert not clip.errors
+    frames = clip.clip_embedding_frames.resolve()
+    # 2 fps -> 5; doubled to 4 fps -> 9 >= 8 satisfies the guarantee
+    assert len(frames) >= 8
