@@ -169,6 +169,177 @@ __global__ __launch_bounds__(256, 2) void k_attn_small(
   }
 }
 
+
+// ---- mid-sequence variant (64 < seq <= 288): ViT-L/14's seq=257 ----
+// Same one-workgroup-per-(frame,head) design: K and V^T stay LDS-
+// resident for the whole sequence (seq<=288 -> ~124 KB with Q/P tiles,
+// 1 WG/CU), Q is processed in 64-row tiles, full softmax per row (no
+// online rescaling needed since all scores are in registers).
+constexpr int SMID = 288;       // padded sequence (18 x 16-col frags)
+constexpr int LDM = SMID + 8;   // LDS s-stride for Vt / P rows
+
+__global__ __launch_bounds__(256, 1) void k_attn_mid(
+    const __bf16* __restrict__ qkv, __bf16* __restrict__ out, long n_frames,
+    int seq, int heads, int hidden, float scale) {
+  const long fh = blockIdx.x;
+  const long frame = fh / heads;
+  const int head = fh % heads;
+  if (frame >= n_frames) return;
+
+  __shared__ __bf16 lds[SMID * LD + 64 * LDM + 64 * LDM + 64 * LD];
+  __bf16* K = lds;                       // [SMID][LD]   rows s, cols d
+  __bf16* Vt = lds + SMID * LD;          // [64][LDM]    rows d, cols s
+  __bf16* P = Vt + 64 * LDM;             // [64][LDM]    rows q, cols s
+  __bf16* Q = P + 64 * LDM;              // [64][LD]     rows q, cols d
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+
+  // ---- K / V^T load: 64 rows per pass, thread t -> row t>>2, d (t&3)*16
+  {
+    const int row = tid >> 2;
+    const int d0 = (tid & 3) * 16;
+#pragma unroll
+    for (int j = 0; j < SMID / 64 + 1; j++) {
+      const int r = j * 64 + row;
+      if (r >= SMID) break;
+      bf16x8 z = {};
+      bf16x8 k0 = z, k1 = z, v0 = z, v1 = z;
+      if (r < seq) {
+        const long base =
+            ((frame * seq + r) * 3) * (long)hidden + (long)head * HD;
+        k0 = *(const bf16x8*)(qkv + base + hidden + d0);
+        k1 = *(const bf16x8*)(qkv + base + hidden + d0 + 8);
+        v0 = *(const bf16x8*)(qkv + base + 2 * hidden + d0);
+        v1 = *(const bf16x8*)(qkv + base + 2 * hidden + d0 + 8);
+      }
+      *(bf16x8*)(K + r * LD + d0) = k0;
+      *(bf16x8*)(K + r * LD + d0 + 8) = k1;
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        Vt[(d0 + e) * LDM + r] = v0[e];
+        Vt[(d0 + 8 + e) * LDM + r] = v1[e];
+      }
+    }
+  }
+
+  const int n_qtiles = (seq + 63) / 64;
+  for (int qt = 0; qt < n_qtiles; qt++) {
+    const int q0 = qt * 64;
+    __syncthreads();  // K/V ready (first iter); Q/P free (later iters)
+    // ---- Q tile load
+    {
+      const int row = tid >> 2;
+      const int d0 = (tid & 3) * 16;
+      bf16x8 z = {};
+      bf16x8 a = z, b = z;
+      if (q0 + row < seq) {
+        const long base =
+            ((frame * seq + q0 + row) * 3) * (long)hidden + (long)head * HD;
+        a = *(const bf16x8*)(qkv + base + d0);
+        b = *(const bf16x8*)(qkv + base + d0 + 8);
+      }
+      *(bf16x8*)(Q + row * LD + d0) = a;
+      *(bf16x8*)(Q + row * LD + d0 + 8) = b;
+    }
+    __syncthreads();
+
+    // ---- S = Q K^T: wave rows [16*wid, +16), 18 col frags
+    f32x4 acc[SMID / 16];
+#pragma unroll
+    for (int n = 0; n < SMID / 16; n++) acc[n] = f32x4{};
+    {
+      const int qrow = 16 * wid + (lane & 15);
+      const int k0e = 8 * (lane >> 4);
+#pragma unroll
+      for (int kk = 0; kk < HD; kk += 32) {
+        bf16x8 qf = *(const bf16x8*)(Q + qrow * LD + kk + k0e);
+#pragma unroll
+        for (int n = 0; n < SMID / 16; n++) {
+          bf16x8 kf =
+              *(const bf16x8*)(K + (n * 16 + (lane & 15)) * LD + kk + k0e);
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc[n], 0, 0, 0);
+        }
+      }
+    }
+
+    // ---- softmax per row (cols masked at seq)
+    {
+      const int colr = lane & 15;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        float mx = -1e30f;
+        float sv[SMID / 16];
+#pragma unroll
+        for (int nn = 0; nn < SMID / 16; nn++) {
+          float sc = acc[nn][reg] * scale;
+          if (nn * 16 + colr >= seq) sc = -1e30f;
+          sv[nn] = sc;
+          mx = fmaxf(mx, sc);
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off));
+        float sum = 0.0f;
+#pragma unroll
+        for (int nn = 0; nn < SMID / 16; nn++) {
+          float e = __expf(sv[nn] - mx);
+          if (nn * 16 + colr >= seq) e = 0.0f;
+          sv[nn] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
+        const float inv = 1.0f / sum;
+#pragma unroll
+        for (int nn = 0; nn < SMID / 16; nn++) acc[nn][reg] = sv[nn] * inv;
+      }
+    }
+    // ---- P to LDS (bf16)
+    {
+      const int colr = lane & 15;
+      const int rbase = 16 * wid + 4 * (lane >> 4);
+#pragma unroll
+      for (int nn = 0; nn < SMID / 16; nn++)
+#pragma unroll
+        for (int reg = 0; reg < 4; reg++)
+          P[(rbase + reg) * LDM + nn * 16 + colr] = (__bf16)acc[nn][reg];
+    }
+    __syncthreads();
+
+    // ---- O = P V : rows [16*wid,+16), d cols 0..63
+    f32x4 oacc[4] = {};
+    {
+      const int prow = 16 * wid + (lane & 15);
+      const int j0 = 8 * (lane >> 4);
+#pragma unroll
+      for (int kk = 0; kk < SMID; kk += 32) {
+        bf16x8 pf = *(const bf16x8*)(P + prow * LDM + kk + j0);
+#pragma unroll
+        for (int n = 0; n < 4; n++) {
+          bf16x8 vf =
+              *(const bf16x8*)(Vt + (n * 16 + (lane & 15)) * LDM + kk + j0);
+          oacc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, oacc[n], 0, 0, 0);
+        }
+      }
+    }
+    {
+      const int dcol = lane & 15;
+      const int rbase = 16 * wid + 4 * (lane >> 4);
+#pragma unroll
+      for (int nn = 0; nn < 4; nn++)
+#pragma unroll
+        for (int reg = 0; reg < 4; reg++) {
+          const int row = q0 + rbase + reg;
+          if (row < seq)
+            out[(frame * seq + row) * (long)hidden + head * HD + nn * 16 +
+                dcol] = (__bf16)oacc[nn][reg];
+        }
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" int cc_attn_small(const void* qkv, void* out, int64_t n_frames,
@@ -189,5 +360,26 @@ extern "C" int cc_attn_small(const void* qkv, void* out, int64_t n_frames,
   if (timed) cc::timed_end("attn_small", stream, ev0, ev1);
   if (e != hipSuccess)
     return cc::set_error(CC_ERR_HIP, "attn launch: %s", hipGetErrorString(e));
+  return CC_OK;
+}
+
+extern "C" int cc_attn_mid(const void* qkv, void* out, int64_t n_frames,
+                           int seq, int heads, int hidden, float scale,
+                           uint64_t stream) {
+  if (!qkv || !out || n_frames <= 0 || seq <= 0)
+    return cc::set_error(CC_ERR_INVALID, "bad attn args");
+  if (seq <= SMAX || seq > SMID || hidden != heads * HD)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "cc_attn_mid needs 64 < seq <= 288 and hd == 64");
+  dim3 block(256), grid((unsigned)(n_frames * heads));
+  hipEvent_t ev0, ev1;
+  bool timed = cc::timed_begin(stream, &ev0, &ev1);
+  hipLaunchKernelGGL(k_attn_mid, grid, block, 0, (hipStream_t)stream,
+                     (const __bf16*)qkv, (__bf16*)out, (long)n_frames, seq,
+                     heads, hidden, scale);
+  hipError_t e = hipGetLastError();
+  if (timed) cc::timed_end("attn_mid", stream, ev0, ev1);
+  if (e != hipSuccess)
+    return cc::set_error(CC_ERR_HIP, "attn_mid launch: %s", hipGetErrorString(e));
   return CC_OK;
 }

@@ -201,7 +201,7 @@ class ClipVisionTowerAMD(torch.nn.Module):
                 getattr(self, f"w_qkv_{i}"),
                 getattr(self, f"b_qkv_{i}"),
             )
-            if qkv_flat.is_cuda and seq <= 64 and hd == 64:
+            if qkv_flat.is_cuda and seq <= 288 and hd == 64:
                 # fused LDS-resident attention straight off the QKV GEMM
                 # output (csrc/cc_attn.hip) — no permute copies
                 lib = hotpath.require_gpu()
@@ -210,8 +210,9 @@ class ClipVisionTowerAMD(torch.nn.Module):
                     device=qkv_flat.device,
                 )
                 stream = torch.cuda.current_stream(qkv_flat.device).cuda_stream
+                attn_fn = lib.cc_attn_small if seq <= 64 else lib.cc_attn_mid
                 hotpath.check(
-                    lib.cc_attn_small(
+                    attn_fn(
                         qkv_flat.data_ptr(), attn.data_ptr(), n, seq,
                         self.heads, cfg.hidden, ctypes.c_float(self.scale),
                         stream,

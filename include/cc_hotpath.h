@@ -163,6 +163,12 @@ int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
 int cc_attn_small(const void* qkv, void* out, int64_t n_frames, int seq,
                   int heads, int hidden, float scale, uint64_t stream);
 
+/* mid-sequence variant (64 < seq <= 288, head dim 64): ViT-L/14's
+ * seq=257.  K/V^T LDS-resident per (frame, head) workgroup, Q in 64-row
+ * tiles (replaces torch sdpa on that path). */
+int cc_attn_mid(const void* qkv, void* out, int64_t n_frames, int seq,
+                int heads, int hidden, float scale, uint64_t stream);
+
 /* ---- fused bf16 LayerNorm (replaces torch layer_norm in the ViT
  * forward; f32 stats/affine, H multiple of 256). */
 int cc_layernorm_bf16(const void* x, const void* w, const void* b, void* y,

@@ -279,3 +279,54 @@ def test_embed_assemble_ln_vs_torch(lib):
     want = torch.nn.functional.layer_norm(
         h.float(), (H,), w, b, eps=1e-5).reshape(n * tokens, H)
     torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
+
+
+def test_attn_mid_vs_sdpa(lib):
+    """cc_attn_mid (seq=257, L/14 shape) == torch sdpa on the same QKV."""
+    import ctypes
+
+    torch.manual_seed(23)
+    n, seq, heads, hd = 3, 257, 16, 64
+    hidden = heads * hd
+    qkv = (torch.randn(n * seq, 3 * hidden) * 0.5).to(torch.bfloat16).cuda()
+    out = torch.empty(n * seq, hidden, dtype=torch.bfloat16, device="cuda")
+    scale = hd ** -0.5
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(lib.cc_attn_mid(
+        qkv.data_ptr(), out.data_ptr(), n, seq, heads, hidden,
+        ctypes.c_float(scale), stream))
+    torch.cuda.synchronize()
+
+    q3 = qkv.reshape(n, seq, 3, heads, hd)
+    q = q3[:, :, 0].permute(0, 2, 1, 3).float()
+    k = q3[:, :, 1].permute(0, 2, 1, 3).float()
+    v = q3[:, :, 2].permute(0, 2, 1, 3).float()
+    want = torch.nn.functional.scaled_dot_product_attention(q, k, v, scale=scale)
+    want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
+    torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
+
+
+def test_attn_mid_ragged_seqs(lib):
+    """Mask correctness at non-multiple-of-64 sequence lengths."""
+    import ctypes
+
+    for seq in [65, 100, 288]:
+        torch.manual_seed(seq)
+        n, heads, hd = 2, 4, 64
+        hidden = heads * hd
+        qkv = (torch.randn(n * seq, 3 * hidden) * 0.5).to(torch.bfloat16).cuda()
+        out = torch.empty(n * seq, hidden, dtype=torch.bfloat16, device="cuda")
+        scale = hd ** -0.5
+        stream = torch.cuda.current_stream().cuda_stream
+        hotpath.check(lib.cc_attn_mid(
+            qkv.data_ptr(), out.data_ptr(), n, seq, heads, hidden,
+            ctypes.c_float(scale), stream))
+        torch.cuda.synchronize()
+        q3 = qkv.reshape(n, seq, 3, heads, hd)
+        q = q3[:, :, 0].permute(0, 2, 1, 3).float()
+        k = q3[:, :, 1].permute(0, 2, 1, 3).float()
+        v = q3[:, :, 2].permute(0, 2, 1, 3).float()
+        want = torch.nn.functional.scaled_dot_product_attention(q, k, v, scale=scale)
+        want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
+        torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2,
+                                   msg=f"seq={seq}")
