@@ -171,14 +171,17 @@ __global__ __launch_bounds__(256, 2) void k_attn_small(
 
 
 // ---- mid-sequence variant (64 < seq <= 288): ViT-L/14's seq=257 ----
-// Same one-workgroup-per-(frame,head) design: K and V^T stay LDS-
-// resident for the whole sequence (seq<=288 -> ~124 KB with Q/P tiles,
-// 1 WG/CU), Q is processed in 64-row tiles, full softmax per row (no
-// online rescaling needed since all scores are in registers).
+// One workgroup per (frame, head), Q in 64-row tiles.  Q and K
+// fragments are 16-byte contiguous in the fused-QKV layout, so they
+// load STRAIGHT from global/L2 (no LDS staging) — only V^T (transposed
+// image for the PV matmul) and P live in LDS: 74 KB -> 2 WG/CU, which
+// doubles the waves/SIMD vs the LDS-everything form (measured 621 us
+// -> see profiles/r01_attn_mid2; the LDS-everything form capped at
+// ~204 TF with 1 wave/SIMD).
 constexpr int SMID = 288;       // padded sequence (18 x 16-col frags)
 constexpr int LDM = SMID + 8;   // LDS s-stride for Vt / P rows
 
-__global__ __launch_bounds__(256, 1) void k_attn_mid(
+__global__ __launch_bounds__(256, 2) void k_attn_mid(
     const __bf16* __restrict__ qkv, __bf16* __restrict__ out, long n_frames,
     int seq, int heads, int hidden, float scale) {
   const long fh = blockIdx.x;
@@ -186,17 +189,16 @@ __global__ __launch_bounds__(256, 1) void k_attn_mid(
   const int head = fh % heads;
   if (frame >= n_frames) return;
 
-  __shared__ __bf16 lds[SMID * LD + 64 * LDM + 64 * LDM + 64 * LD];
-  __bf16* K = lds;                       // [SMID][LD]   rows s, cols d
-  __bf16* Vt = lds + SMID * LD;          // [64][LDM]    rows d, cols s
-  __bf16* P = Vt + 64 * LDM;             // [64][LDM]    rows q, cols s
-  __bf16* Q = P + 64 * LDM;              // [64][LD]     rows q, cols d
+  __shared__ __bf16 lds[2 * 64 * LDM];
+  __bf16* Vt = lds;              // [64][LDM] rows d, cols s
+  __bf16* P = lds + 64 * LDM;    // [64][LDM] rows q, cols s
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
+  const long qkv_base = (frame * (long)seq * 3) * hidden + (long)head * HD;
 
-  // ---- K / V^T load: 64 rows per pass, thread t -> row t>>2, d (t&3)*16
+  // ---- V^T load: 64 rows per pass, thread t -> row t>>2, d (t&3)*16
   {
     const int row = tid >> 2;
     const int d0 = (tid & 3) * 16;
@@ -205,17 +207,12 @@ __global__ __launch_bounds__(256, 1) void k_attn_mid(
       const int r = j * 64 + row;
       if (r >= SMID) break;
       bf16x8 z = {};
-      bf16x8 k0 = z, k1 = z, v0 = z, v1 = z;
+      bf16x8 v0 = z, v1 = z;
       if (r < seq) {
-        const long base =
-            ((frame * seq + r) * 3) * (long)hidden + (long)head * HD;
-        k0 = *(const bf16x8*)(qkv + base + hidden + d0);
-        k1 = *(const bf16x8*)(qkv + base + hidden + d0 + 8);
-        v0 = *(const bf16x8*)(qkv + base + 2 * hidden + d0);
-        v1 = *(const bf16x8*)(qkv + base + 2 * hidden + d0 + 8);
+        const long base = qkv_base + (long)r * 3 * hidden + 2 * hidden;
+        v0 = *(const bf16x8*)(qkv + base + d0);
+        v1 = *(const bf16x8*)(qkv + base + d0 + 8);
       }
-      *(bf16x8*)(K + r * LD + d0) = k0;
-      *(bf16x8*)(K + r * LD + d0 + 8) = k1;
 #pragma unroll
       for (int e = 0; e < 8; e++) {
         Vt[(d0 + e) * LDM + r] = v0[e];
@@ -227,38 +224,28 @@ __global__ __launch_bounds__(256, 1) void k_attn_mid(
   const int n_qtiles = (seq + 63) / 64;
   for (int qt = 0; qt < n_qtiles; qt++) {
     const int q0 = qt * 64;
-    __syncthreads();  // K/V ready (first iter); Q/P free (later iters)
-    // ---- Q tile load
-    {
-      const int row = tid >> 2;
-      const int d0 = (tid & 3) * 16;
-      bf16x8 z = {};
-      bf16x8 a = z, b = z;
-      if (q0 + row < seq) {
-        const long base =
-            ((frame * seq + q0 + row) * 3) * (long)hidden + (long)head * HD;
-        a = *(const bf16x8*)(qkv + base + d0);
-        b = *(const bf16x8*)(qkv + base + d0 + 8);
-      }
-      *(bf16x8*)(Q + row * LD + d0) = a;
-      *(bf16x8*)(Q + row * LD + d0 + 8) = b;
-    }
-    __syncthreads();
+    __syncthreads();  // Vt ready (first iter); P free (later iters)
 
-    // ---- S = Q K^T: wave rows [16*wid, +16), 18 col frags
+    // ---- S = Q K^T: wave rows [16*wid, +16), 18 col frags; Q/K frags
+    // read directly from global (16B contiguous), rows clamped to seq-1
+    // (masked in softmax anyway).
     f32x4 acc[SMID / 16];
 #pragma unroll
     for (int n = 0; n < SMID / 16; n++) acc[n] = f32x4{};
     {
-      const int qrow = 16 * wid + (lane & 15);
+      int qrow = q0 + 16 * wid + (lane & 15);
+      qrow = qrow < seq ? qrow : seq - 1;
       const int k0e = 8 * (lane >> 4);
 #pragma unroll
       for (int kk = 0; kk < HD; kk += 32) {
-        bf16x8 qf = *(const bf16x8*)(Q + qrow * LD + kk + k0e);
+        bf16x8 qf =
+            *(const bf16x8*)(qkv + qkv_base + (long)qrow * 3 * hidden + kk + k0e);
 #pragma unroll
         for (int n = 0; n < SMID / 16; n++) {
-          bf16x8 kf =
-              *(const bf16x8*)(K + (n * 16 + (lane & 15)) * LD + kk + k0e);
+          int krow = n * 16 + (lane & 15);
+          krow = krow < seq ? krow : seq - 1;
+          bf16x8 kf = *(const bf16x8*)(qkv + qkv_base + (long)krow * 3 * hidden +
+                                       hidden + kk + k0e);
           acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc[n], 0, 0, 0);
         }
       }
