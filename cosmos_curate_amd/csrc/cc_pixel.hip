@@ -256,31 +256,65 @@ __global__ void k_clip_preprocess_f32(const unsigned char* __restrict__ in,
   }
 }
 
+__device__ __forceinline__ unsigned short pp_bf16(float x, float mean,
+                                                  float stdev) {
+  float v = (x - mean) / stdev;  // division kept: bit-exact vs oracle
+  // round-to-nearest-even bf16 (torch .to(bfloat16) semantics)
+  union {
+    float f;
+    unsigned int u;
+  } cv{v};
+  unsigned int lsb = (cv.u >> 16) & 1;
+  unsigned int r = cv.u + 0x7fffu + lsb;
+  return (unsigned short)(r >> 16);
+}
+
+typedef __attribute__((ext_vector_type(4))) unsigned short u16x4;
+
+// 4 pixels per thread: 3 dword reads of the interleaved u8 RGB, one
+// 8-byte store per plane (coalesced both ways); scalar tail for
+// hw % 4 != 0 pixels.
 __global__ void k_clip_preprocess_bf16(const unsigned char* __restrict__ in,
                                        int n, int h, int w, float m0, float m1,
                                        float m2, float s0, float s1, float s2,
                                        unsigned short* __restrict__ out) {
-  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;  // quad index
   long hw = (long)h * w;
-  long total = (long)n * hw;
-  if (idx >= total) return;
-  long p = idx % hw;
-  int f = idx / hw;
-  const unsigned char* px = in + ((size_t)f * hw + p) * 3;
-  unsigned short* ob = out + (size_t)f * 3 * hw + p;
-  float mean[3] = {m0, m1, m2}, stdev[3] = {s0, s1, s2};
+  long total_q = ((long)n * hw + 3) / 4;
+  if (idx >= total_q) return;
+  const float mean[3] = {m0, m1, m2};
+  const float stdev[3] = {s0, s1, s2};
+  long pix0 = idx * 4;
+  long p = pix0 % hw;
+  int f = pix0 / hw;
+  unsigned char bytes[12];
+  if (p + 4 <= hw && ((long)n * hw - pix0) >= 4) {
+    const unsigned int* src = (const unsigned int*)(in + ((size_t)f * hw + p) * 3);
+    unsigned int d0 = src[0], d1 = src[1], d2 = src[2];
+    *(unsigned int*)(bytes + 0) = d0;
+    *(unsigned int*)(bytes + 4) = d1;
+    *(unsigned int*)(bytes + 8) = d2;
+    unsigned short* ob = out + (size_t)f * 3 * hw + p;
 #pragma unroll
-  for (int c = 0; c < 3; c++) {
-    float x = (float)px[c] / 255.0f;
-    float v = (x - mean[c]) / stdev[c];
-    // round-to-nearest-even bf16 (torch .to(bfloat16) semantics)
-    union {
-      float f;
-      unsigned int u;
-    } cv{v};
-    unsigned int lsb = (cv.u >> 16) & 1;
-    unsigned int r = cv.u + 0x7fffu + lsb;
-    ob[(size_t)c * hw] = (unsigned short)(r >> 16);
+    for (int c = 0; c < 3; c++) {
+      u16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        o[j] = pp_bf16((float)bytes[3 * j + c] / 255.0f, mean[c], stdev[c]);
+      *(u16x4*)(ob + (size_t)c * hw) = o;
+    }
+  } else {
+    // tail quad may cross a frame/row boundary: per-pixel path
+    long total = (long)n * hw;
+    for (long px = pix0; px < pix0 + 4 && px < total; px++) {
+      long pp = px % hw;
+      int ff = px / hw;
+      const unsigned char* sp = in + ((size_t)ff * hw + pp) * 3;
+      unsigned short* ob = out + (size_t)ff * 3 * hw + pp;
+#pragma unroll
+      for (int c = 0; c < 3; c++)
+        ob[(size_t)c * hw] = pp_bf16((float)sp[c] / 255.0f, mean[c], stdev[c]);
+    }
   }
 }
 
@@ -454,7 +488,8 @@ int cc_clip_preprocess(const void* in, int n, int h, int w, const float mean[3],
               (const unsigned char*)in, n, h, w, mean[0], mean[1], mean[2],
               stdev[0], stdev[1], stdev[2], (float*)out);
   } else if (out_dtype == 1) {
-    CC_LAUNCH("clip_preprocess", grid, block, stream, k_clip_preprocess_bf16,
+    dim3 gridq(((total + 3) / 4 + 255) / 256);  // 4 pixels per thread
+    CC_LAUNCH("clip_preprocess", gridq, block, stream, k_clip_preprocess_bf16,
               (const unsigned char*)in, n, h, w, mean[0], mean[1], mean[2],
               stdev[0], stdev[1], stdev[2], (unsigned short*)out);
   } else {
