@@ -29,38 +29,51 @@ constexpr int SMAX = 64;   // padded sequence tile
 constexpr int HD = 64;     // head dim
 constexpr int LD = 72;     // LDS row stride (pad 8 elements vs 64 banks)
 
-__global__ __launch_bounds__(256, 4) void k_attn_small(
+// NOTE: the direct-global Q/K variant (4 WG/CU) was tried here too and
+// measured SLOWER (89 -> 126 us/launch, profiles/r01_rocprof_b32_afterfusions.txt)
+// — at seq=50 the whole K tile L1/LDS-fits and re-reading it from L2
+// per wave costs more than the staging saves (guide rule 7); the
+// LDS-resident form below stays.  k_attn_mid (seq 257) keeps the
+// direct-global form where it measured FASTER (L/14 step 80.3 -> 75.1 ms).
+__global__ __launch_bounds__(256, 2) void k_attn_small(
     const __bf16* __restrict__ qkv, __bf16* __restrict__ out, long n_frames,
     int seq, int heads, int hidden, float scale) {
-  // one workgroup per (frame, head).  Q/K fragments are 16B-contiguous
-  // in the fused-QKV layout -> loaded straight from global/L2 (same
-  // trick as k_attn_mid below); LDS holds only V^T and P (37 KB ->
-  // 4 WG/CU).
+  // one workgroup per (frame, head)
   const long fh = blockIdx.x;
   const long frame = fh / heads;
   const int head = fh % heads;
   if (frame >= n_frames) return;
 
-  __shared__ __bf16 lds[2 * SMAX * LD];
-  __bf16* Vt = lds;               // [64][LD] rows d, cols s
-  __bf16* P = lds + SMAX * LD;    // [64][LD] rows q, cols s
+  __shared__ __bf16 lds[4 * SMAX * LD];  // Q, K, Vt, P
+  __bf16* Q = lds;
+  __bf16* K = lds + SMAX * LD;
+  __bf16* Vt = lds + 2 * SMAX * LD;
+  __bf16* P = lds + 3 * SMAX * LD;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;  // 4 waves; wave w owns S rows [16w, 16w+16)
-  const long qkv_base = (frame * (long)seq * 3) * hidden + (long)head * HD;
 
-  // ---- V^T: thread t handles row t>>2, cols (t&3)*16.. ----
+  // ---- load Q/K/V tiles: thread t handles row t>>2, cols (t&3)*16.. ----
   {
     const int row = tid >> 2;
     const int d0 = (tid & 3) * 16;
+    const long base = ((frame * seq + row) * 3) * (long)hidden + (long)head * HD;
     bf16x8 z = {};
-    bf16x8 v0 = z, v1 = z;
+    bf16x8 q0 = z, q1 = z, k0 = z, k1 = z, v0 = z, v1 = z;
     if (row < seq) {
-      const long base = qkv_base + (long)row * 3 * hidden + 2 * hidden;
-      v0 = *(const bf16x8*)(qkv + base + d0);
-      v1 = *(const bf16x8*)(qkv + base + d0 + 8);
+      q0 = *(const bf16x8*)(qkv + base + d0);
+      q1 = *(const bf16x8*)(qkv + base + d0 + 8);
+      k0 = *(const bf16x8*)(qkv + base + hidden + d0);
+      k1 = *(const bf16x8*)(qkv + base + hidden + d0 + 8);
+      v0 = *(const bf16x8*)(qkv + base + 2 * hidden + d0);
+      v1 = *(const bf16x8*)(qkv + base + 2 * hidden + d0 + 8);
     }
+    *(bf16x8*)(Q + row * LD + d0) = q0;
+    *(bf16x8*)(Q + row * LD + d0 + 8) = q1;
+    *(bf16x8*)(K + row * LD + d0) = k0;
+    *(bf16x8*)(K + row * LD + d0 + 8) = k1;
+    // V transposed: Vt[d][row] = V[row][d]
 #pragma unroll
     for (int e = 0; e < 8; e++) {
       Vt[(d0 + e) * LD + row] = v0[e];
@@ -72,19 +85,14 @@ __global__ __launch_bounds__(256, 4) void k_attn_small(
   // ---- S = Q K^T * scale, rows [16w,16w+16), cols 0..63 ----
   f32x4 acc[4] = {};
   {
-    int qrow = 16 * wid + (lane & 15);
-    qrow = qrow < seq ? qrow : seq - 1;  // masked rows: any valid data
+    const int qrow = 16 * wid + (lane & 15);
     const int k0e = 8 * (lane >> 4);
 #pragma unroll
     for (int kk = 0; kk < HD; kk += 32) {
-      bf16x8 qf =
-          *(const bf16x8*)(qkv + qkv_base + (long)qrow * 3 * hidden + kk + k0e);
+      bf16x8 qf = *(const bf16x8*)(Q + qrow * LD + kk + k0e);
 #pragma unroll
       for (int n = 0; n < 4; n++) {
-        int krow = n * 16 + (lane & 15);
-        krow = krow < seq ? krow : seq - 1;
-        bf16x8 kf = *(const bf16x8*)(qkv + qkv_base + (long)krow * 3 * hidden +
-                                     hidden + kk + k0e);
+        bf16x8 kf = *(const bf16x8*)(K + (n * 16 + (lane & 15)) * LD + kk + k0e);
         acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc[n], 0, 0, 0);
       }
     }
@@ -166,6 +174,7 @@ __global__ __launch_bounds__(256, 4) void k_attn_small(
       }
   }
 }
+
 
 // ---- mid-sequence variant (64 < seq <= 288): ViT-L/14's seq=257 ----
 // One workgroup per (frame, head), Q in 64-row tiles.  Q and K
