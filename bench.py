@@ -167,8 +167,9 @@ def main() -> None:
             )
         )
         if args.streams == 1:
-            pixels = model.preprocess_u8(rgb)
-            emb = model.tower(pixels)  # (F, proj) f32 unit-norm
+            # fused normalize+patch-extract straight into the GEMM layout
+            patches = model.preprocess_patches_u8(rgb)
+            emb = model.tower(patches=patches, n=F)  # (F, proj) f32 unit-norm
         else:
             # split the batch across streams: one half's bandwidth-bound
             # kernels overlap the other half's MFMA chain
@@ -180,7 +181,9 @@ def main() -> None:
                 with torch.cuda.stream(st):
                     if st is not main:
                         st.wait_stream(main)
-                    parts.append(model.tower(model.preprocess_u8(chunk)))
+                    parts.append(model.tower(
+                        patches=model.preprocess_patches_u8(chunk),
+                        n=chunk.shape[0]))
             for st in side_streams:
                 main.wait_stream(st)
             emb = torch.cat(parts, dim=0)

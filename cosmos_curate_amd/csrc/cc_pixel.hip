@@ -318,6 +318,59 @@ __global__ void k_clip_preprocess_bf16(const unsigned char* __restrict__ in,
   }
 }
 
+
+// normalize + patch-extraction fused: u8 NHWC frames -> the ViT patch
+// GEMM's A layout [n*g*g, kpad] bf16, col = c*P*P + ky*P + kx (the
+// conv-weight flatten order), cols >= 3*P*P zero-padded.  Replaces the
+// torch reshape/permute/pad chain after cc_clip_preprocess (a full
+// 2x206 MB round trip per bench step).
+__global__ void k_clip_preprocess_patches(
+    const unsigned char* __restrict__ in, int n, int h, int w, int patch,
+    int kpad, float m0, float m1, float m2, float s0, float s1, float s2,
+    unsigned short* __restrict__ out) {
+  const int g = w / patch;
+  const int pp2 = patch * patch;
+  const long rows = (long)n * (h / patch) * g;
+  // quad of consecutive cols
+  long q = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long quads_per_row = kpad / 4;
+  const long row = q / quads_per_row;
+  if (row >= rows) return;
+  const int col0 = (int)(q - row * quads_per_row) * 4;
+  const float mean[3] = {m0, m1, m2};
+  const float stdev[3] = {s0, s1, s2};
+  const long f = row / ((h / patch) * g);
+  const int pr = (int)(row - f * (h / patch) * g);
+  const int ph = pr / g, pw = pr % g;
+  u16x4 o = {0, 0, 0, 0};
+  const bool quad_fast = (patch % 4 == 0) && (col0 + 4 <= 3 * pp2);
+  if (quad_fast) {
+    const int c = col0 / pp2;
+    const int r = col0 - c * pp2;
+    const int ky = r / patch, kx = r % patch;  // kx % 4 == 0 when P%4==0
+    const long y = (long)ph * patch + ky;
+    const long x = (long)pw * patch + kx;
+    const unsigned char* px = in + (((size_t)f * h + y) * w + x) * 3;
+#pragma unroll
+    for (int j = 0; j < 4; j++)
+      o[j] = pp_bf16((float)px[3 * j + c] / 255.0f, mean[c], stdev[c]);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 4; j++) {
+      const int col = col0 + j;
+      if (col >= 3 * pp2) break;  // zero pad
+      const int c = col / pp2;
+      const int r = col - c * pp2;
+      const int ky = r / patch, kx = r % patch;
+      const long y = (long)ph * patch + ky;
+      const long x = (long)pw * patch + kx;
+      o[j] = pp_bf16((float)in[(((size_t)f * h + y) * w + x) * 3 + c] / 255.0f,
+                     mean[c], stdev[c]);
+    }
+  }
+  *(u16x4*)(out + row * kpad + col0) = o;
+}
+
 // ---------------- gather + duplicate broadcast ----------------
 __global__ void k_gather_frames_u8(const unsigned char* __restrict__ frames,
                                    size_t frame_bytes,
@@ -495,6 +548,26 @@ int cc_clip_preprocess(const void* in, int n, int h, int w, const float mean[3],
   } else {
     return cc::set_error(CC_ERR_INVALID, "out_dtype must be 0(f32)|1(bf16)");
   }
+  return CC_OK;
+}
+
+
+int cc_clip_preprocess_patches(const void* in, int n, int h, int w, int patch,
+                               int kpad, const float mean[3],
+                               const float stdev[3], void* out,
+                               uint64_t stream) {
+  if (!in || !out || !mean || !stdev || n <= 0 || patch <= 0)
+    return cc::set_error(CC_ERR_INVALID, "bad args");
+  if (h % patch != 0 || w % patch != 0 || kpad % 64 != 0 ||
+      kpad < 3 * patch * patch)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "need h,w %% patch == 0 and kpad %% 64 == 0, >= 3*P*P");
+  long rows = (long)n * (h / patch) * (w / patch);
+  long quads = rows * (kpad / 4);
+  dim3 block(256), grid((quads + 255) / 256);
+  CC_LAUNCH("clip_preprocess", grid, block, stream, k_clip_preprocess_patches,
+            (const unsigned char*)in, n, h, w, patch, kpad, mean[0], mean[1],
+            mean[2], stdev[0], stdev[1], stdev[2], (unsigned short*)out);
   return CC_OK;
 }
 

@@ -61,6 +61,9 @@ def _declare(lib: ctypes.CDLL) -> None:
     lib.cc_clip_preprocess.argtypes = [
         c.c_void_p, c.c_int, c.c_int, c.c_int, c.POINTER(c.c_float),
         c.POINTER(c.c_float), c.c_void_p, c.c_int, c.c_uint64]
+    lib.cc_clip_preprocess_patches.argtypes = [
+        c.c_void_p, c.c_int, c.c_int, c.c_int, c.c_int, c.c_int,
+        c.POINTER(c.c_float), c.POINTER(c.c_float), c.c_void_p, c.c_uint64]
     lib.cc_gather_frames_u8.argtypes = [
         c.c_void_p, c.c_int, c.c_size_t, c.c_void_p, c.c_void_p, c.c_int,
         c.c_int, c.c_void_p, c.c_uint64]

@@ -65,16 +65,47 @@ class _CLIPImageEmbeddings(torch.nn.Module):
         )
         return out
 
+    def preprocess_patches_u8(self, frames_dev_u8: torch.Tensor) -> torch.Tensor:
+        """(N,224,224,3) u8 on device -> GEMM-ready patch rows
+        [N*g*g, patch_k] bf16 in one fused kernel (normalize +
+        patch-extract; replaces preprocess_u8 + the tower's torch
+        reshape/permute/pad)."""
+        import ctypes
+
+        lib = hotpath.require_gpu()
+        n, h, w, _ = frames_dev_u8.shape
+        cfg = self.tower.cfg
+        g = cfg.image // cfg.patch
+        out = torch.empty(
+            (n * g * g, self.tower.patch_k), dtype=torch.bfloat16,
+            device=frames_dev_u8.device,
+        )
+        stream = torch.cuda.current_stream(frames_dev_u8.device).cuda_stream
+        mean = (ctypes.c_float * 3)(*self._mean_arr)
+        std = (ctypes.c_float * 3)(*self._std_arr)
+        hotpath.check(
+            lib.cc_clip_preprocess_patches(
+                frames_dev_u8.contiguous().data_ptr(), n, h, w, cfg.patch,
+                self.tower.patch_k, mean, std, out.data_ptr(), stream,
+            )
+        )
+        return out
+
     @torch.no_grad()
     def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
         if isinstance(images, np.ndarray):
             # (N,H,W,C) u8 host array (reference clip.py:66-68 entry form)
-            dev = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
-            pixels = self.preprocess_u8(dev)
-        elif images.dtype == torch.uint8 and images.ndim == 4 and images.shape[-1] == 3:
-            pixels = self.preprocess_u8(images.to(self.device))
-        else:
-            pixels = images.to(self.device, dtype=torch.bfloat16)
+            images = torch.from_numpy(np.ascontiguousarray(images)).to(self.device)
+        if (
+            images.dtype == torch.uint8 and images.ndim == 4
+            and images.shape[-1] == 3
+            and images.shape[1] == self.tower.cfg.image
+            and images.shape[2] == self.tower.cfg.image
+        ):
+            n = images.shape[0]
+            patches = self.preprocess_patches_u8(images.to(self.device))
+            return self.tower(patches=patches, n=n)
+        pixels = images.to(self.device, dtype=torch.bfloat16)
         return self.tower(pixels)
 
 

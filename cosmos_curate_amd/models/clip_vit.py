@@ -156,21 +156,35 @@ class ClipVisionTowerAMD(torch.nn.Module):
         ).to(x.dtype)
 
     @torch.no_grad()
-    def forward(self, pixel_values: torch.Tensor) -> torch.Tensor:
-        """(N,3,224,224) bf16 -> (N,proj) f32 L2-normalized embeddings."""
+    def forward(
+        self,
+        pixel_values: torch.Tensor | None = None,
+        *,
+        patches: torch.Tensor | None = None,
+        n: int | None = None,
+    ) -> torch.Tensor:
+        """(N,3,224,224) bf16 -> (N,proj) f32 L2-normalized embeddings.
+
+        Alternatively accepts prebuilt GEMM-ready ``patches``
+        [N*g*g, patch_k] bf16 (cc_clip_preprocess_patches output) with
+        ``n`` = N, skipping the torch reshape/permute/pad chain.
+        """
         cfg = self.cfg
-        x = pixel_values.to(torch.bfloat16)
-        n = x.shape[0]
         g = cfg.image // cfg.patch
         k0 = 3 * cfg.patch * cfg.patch
-        # patch extraction: (n,c,ph,ky,pw,kx) -> (n,ph,pw,c,ky,kx)
-        patches = (
-            x.reshape(n, 3, g, cfg.patch, g, cfg.patch)
-            .permute(0, 2, 4, 1, 3, 5)
-            .reshape(n * g * g, k0)
-        )
-        if self.patch_k != k0:
-            patches = torch.nn.functional.pad(patches, (0, self.patch_k - k0))
+        if patches is None:
+            x = pixel_values.to(torch.bfloat16)
+            n = x.shape[0]
+            # patch extraction: (n,c,ph,ky,pw,kx) -> (n,ph,pw,c,ky,kx)
+            patches = (
+                x.reshape(n, 3, g, cfg.patch, g, cfg.patch)
+                .permute(0, 2, 4, 1, 3, 5)
+                .reshape(n * g * g, k0)
+            )
+            if self.patch_k != k0:
+                patches = torch.nn.functional.pad(patches, (0, self.patch_k - k0))
+        else:
+            assert n is not None and patches.shape == (n * g * g, self.patch_k)
         tok_flat = self._linear(patches, self.w_patch, None)  # (n*g*g, hidden)
         seq = g * g + 1
         if tok_flat.is_cuda:

@@ -133,6 +133,15 @@ int cc_resize_bicubic_u8(const void* in, int n, int src_h, int src_w,
 int cc_clip_preprocess(const void* in, int n, int h, int w,
                        const float mean[3], const float stdev[3],
                        void* out, int out_dtype, uint64_t stream);
+/* Normalize + patch-extraction fused: u8 NHWC frames -> the ViT patch
+ * GEMM's A layout [n*(h/patch)*(w/patch), kpad] bf16, col order
+ * c*P*P + ky*P + kx (conv-weight flatten), zero-padded to kpad (64-
+ * multiple).  Replaces the torch reshape/permute/pad chain after
+ * cc_clip_preprocess. */
+int cc_clip_preprocess_patches(const void* in, int n, int h, int w, int patch,
+                               int kpad, const float mean[3],
+                               const float stdev[3], void* out,
+                               uint64_t stream);
 /* Gather + duplicate-count broadcast of selected frames on device
  * (the decode loop's count broadcast, decoder_utils.py:447-453). */
 int cc_gather_frames_u8(const void* frames, int n_in, size_t frame_bytes,

@@ -330,3 +330,31 @@ def test_attn_mid_ragged_seqs(lib):
         want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
         torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2,
                                    msg=f"seq={seq}")
+
+
+def test_preprocess_patches_matches_torch_chain(lib):
+    """cc_clip_preprocess_patches == cc_clip_preprocess + torch
+    reshape/permute/pad, bit-exact (same normalize arithmetic)."""
+    from cosmos_curate_amd.models.clip import _CLIPImageEmbeddings
+
+    m = _CLIPImageEmbeddings("vit_b32")
+    torch.manual_seed(4)
+    frames = torch.randint(0, 256, (5, 224, 224, 3), dtype=torch.uint8).cuda()
+    fused = m.preprocess_patches_u8(frames)
+
+    pixels = m.preprocess_u8(frames)
+    cfg = m.tower.cfg
+    g = cfg.image // cfg.patch
+    k0 = 3 * cfg.patch * cfg.patch
+    want = (
+        pixels.reshape(5, 3, g, cfg.patch, g, cfg.patch)
+        .permute(0, 2, 4, 1, 3, 5)
+        .reshape(5 * g * g, k0)
+    )
+    if m.tower.patch_k != k0:
+        want = torch.nn.functional.pad(want, (0, m.tower.patch_k - k0))
+    assert torch.equal(fused, want)
+    # and the end-to-end u8 path equals the pixels path
+    e1 = m.tower(patches=fused, n=5)
+    e2 = m.tower(pixels)
+    assert torch.equal(e1, e2)
