@@ -226,13 +226,11 @@ static bool parse_trak(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
 }
 
 static void build_ps_prefix(cc_demux* t) {
-  // AnnexB parameter-set prefix from avcC (h264 only for now)
-  if (t->codec != 0 || t->avcc.size() < 7) return;
+  // AnnexB parameter-set prefix from avcC (h264) or hvcC (hevc)
+  static const uint8_t sc[4] = {0, 0, 0, 1};
   const uint8_t* a = t->avcc.data();
-  t->nal_len_size = (a[4] & 0x03) + 1;
-  size_t pos = 5;
+  size_t pos = 0;
   auto append_sets = [&](int count) {
-    static const uint8_t sc[4] = {0, 0, 0, 1};
     for (int i = 0; i < count && pos + 2 <= t->avcc.size(); i++) {
       uint16_t ln = rd16(a + pos);
       pos += 2;
@@ -242,12 +240,28 @@ static void build_ps_prefix(cc_demux* t) {
       pos += ln;
     }
   };
-  int num_sps = a[5] & 0x1F;
-  pos = 6;  // skip the SPS-count byte
-  append_sets(num_sps);
-  if (pos < t->avcc.size()) {
-    int num_pps = a[pos++];
-    append_sets(num_pps);
+  if (t->codec == 0 && t->avcc.size() >= 7) {
+    t->nal_len_size = (a[4] & 0x03) + 1;
+    int num_sps = a[5] & 0x1F;
+    pos = 6;  // skip the SPS-count byte
+    append_sets(num_sps);
+    if (pos < t->avcc.size()) {
+      int num_pps = a[pos++];
+      append_sets(num_pps);
+    }
+  } else if (t->codec == 1 && t->avcc.size() >= 23) {
+    // HEVCDecoderConfigurationRecord (ISO 14496-15 §8.3.3.1): 22-byte
+    // header (byte 21 low bits = lengthSizeMinusOne), numOfArrays at 22,
+    // then arrays of (1-byte type, 2-byte count, count x (len, nalu)) —
+    // VPS/SPS/PPS all go into the sync-sample prefix.
+    t->nal_len_size = (a[21] & 0x03) + 1;
+    int num_arrays = a[22];
+    pos = 23;
+    for (int ar = 0; ar < num_arrays && pos + 3 <= t->avcc.size(); ar++) {
+      int cnt = rd16(a + pos + 1);
+      pos += 3;
+      append_sets(cnt);
+    }
   }
 }
 

@@ -235,25 +235,40 @@ def annexb_packets(data: bytes, track: Mp4Track) -> list[bytes]:
     Mirrors what PyNvDemuxer hands to NVDEC (nvcodec_utils.py:224); used as
     the golden reference for the C++ host demuxer's packet output.
     """
-    if not track.avcc:
-        msg = "avcC missing (only h264 supported here)"
+    if track.avcc:
+        nal_len = (track.avcc[4] & 0x03) + 1
+        # parameter sets from avcC
+        prefix = b""
+        pos = 5
+        num_sps = track.avcc[pos] & 0x1F
+        pos += 1
+        for _ in range(num_sps):
+            ln = struct.unpack_from(">H", track.avcc, pos)[0]
+            prefix += b"\x00\x00\x00\x01" + track.avcc[pos + 2 : pos + 2 + ln]
+            pos += 2 + ln
+        num_pps = track.avcc[pos]
+        pos += 1
+        for _ in range(num_pps):
+            ln = struct.unpack_from(">H", track.avcc, pos)[0]
+            prefix += b"\x00\x00\x00\x01" + track.avcc[pos + 2 : pos + 2 + ln]
+            pos += 2 + ln
+    elif track.hvcc:
+        # HEVCDecoderConfigurationRecord (ISO 14496-15 §8.3.3.1):
+        # 22-byte header, numOfArrays, arrays of (type, count, nalus)
+        h = track.hvcc
+        nal_len = (h[21] & 0x03) + 1
+        prefix = b""
+        pos = 23
+        for _ in range(h[22]):
+            cnt = struct.unpack_from(">H", h, pos + 1)[0]
+            pos += 3
+            for _ in range(cnt):
+                ln = struct.unpack_from(">H", h, pos)[0]
+                prefix += b"\x00\x00\x00\x01" + h[pos + 2 : pos + 2 + ln]
+                pos += 2 + ln
+    else:
+        msg = "avcC/hvcC missing"
         raise ValueError(msg)
-    nal_len = (track.avcc[4] & 0x03) + 1
-    # parameter sets from avcC
-    prefix = b""
-    pos = 5
-    num_sps = track.avcc[pos] & 0x1F
-    pos += 1
-    for _ in range(num_sps):
-        ln = struct.unpack_from(">H", track.avcc, pos)[0]
-        prefix += b"\x00\x00\x00\x01" + track.avcc[pos + 2 : pos + 2 + ln]
-        pos += 2 + ln
-    num_pps = track.avcc[pos]
-    pos += 1
-    for _ in range(num_pps):
-        ln = struct.unpack_from(">H", track.avcc, pos)[0]
-        prefix += b"\x00\x00\x00\x01" + track.avcc[pos + 2 : pos + 2 + ln]
-        pos += 2 + ln
 
     out: list[bytes] = []
     sync = set(track.sync_samples) if track.sync_samples else None

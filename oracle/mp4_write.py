@@ -31,8 +31,9 @@ def write_mp4(
     width: int = 64,
     height: int = 64,
     payload_byte: int = 0xAB,
+    codec: str = "h264",
 ) -> bytes:
-    """Build an MP4 with one avc1 video track and fake sample data."""
+    """Build an MP4 with one avc1 (or hvc1) video track and fake samples."""
     n = len(sample_sizes)
     mdat_payload = bytes([payload_byte]) * sum(sample_sizes)
     # layout: ftyp, moov, mdat.  Compute moov size by building it with a
@@ -42,8 +43,21 @@ def write_mp4(
     fake_avcc = bytes(
         [1, 0x42, 0xC0, 0x1E, 0xFF, 0xE1, 0, 4, 0x67, 0x42, 0xC0, 0x1E, 1, 0, 2, 0x68, 0xCE]
     )
+    # HEVCDecoderConfigurationRecord: 22-byte header (byte 21 low bits =
+    # lengthSizeMinusOne=3 -> 4-byte NAL lengths), numOfArrays=3, then one
+    # fake VPS(32)/SPS(33)/PPS(34) NAL each (ISO 14496-15 §8.3.3.1)
+    fake_hvcc = (
+        bytes([1]) + bytes(20) + bytes([0x03, 3])
+        + bytes([0xA0, 0, 1, 0, 3, 0x40, 1, 2])       # VPS array, 1 nal len 3
+        + bytes([0xA1, 0, 1, 0, 4, 0x42, 1, 2, 3])    # SPS array, 1 nal len 4
+        + bytes([0xA2, 0, 1, 0, 2, 0x44, 1])          # PPS array, 1 nal len 2
+    )
+    entry_type, cfg_box = (
+        (b"hvc1", _box(b"hvcC", fake_hvcc)) if codec == "hevc"
+        else (b"avc1", _box(b"avcC", fake_avcc))
+    )
     avc1 = _box(
-        b"avc1",
+        entry_type,
         b"\x00" * 6
         + struct.pack(">H", 1)
         + b"\x00" * 16
@@ -54,7 +68,7 @@ def write_mp4(
         + b"\x00" * 32
         + struct.pack(">H", 0x18)
         + struct.pack(">h", -1)
-        + _box(b"avcC", fake_avcc),
+        + cfg_box,
     )
     stsd = _full(b"stsd", 0, 0, struct.pack(">I", 1) + avc1)
     stts_b = _full(
