@@ -434,8 +434,9 @@ void cc_buffer_free(void* p) { free(p); }
 int cc_demux_remux_clip(cc_demux* d, double start_s, double end_s,
                         uint8_t** out, size_t* out_size) {
   if (!d || !out || !out_size) return cc::set_error(CC_ERR_INVALID, "null arg");
-  if (d->codec != 0 || d->avcc.empty())
-    return cc::set_error(CC_ERR_UNSUPPORTED, "remux supports h264/avcC only");
+  if ((d->codec != 0 && d->codec != 1) || d->avcc.empty())
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "remux needs an avcC/hvcC h264 or hevc track");
   const size_t n = d->dts.size();
   if (d->offsets.size() != n || d->sizes.size() != n)
     return cc::set_error(CC_ERR_PARSE, "incomplete sample tables");
@@ -547,14 +548,15 @@ int cc_demux_remux_clip(cc_demux* d, double start_s, double end_s,
             {
               BoxW stsd(b, "stsd");
               b.u32(0); b.u32(1);
-              BoxW avc1(b, "avc1");
+              // entry + config box types follow the source codec
+              BoxW avc1(b, d->codec == 1 ? "hvc1" : "avc1");
               b.zeros(6); b.u16(1);
               b.zeros(16);
               b.u16((uint16_t)d->width); b.u16((uint16_t)d->height);
               b.u32(0x00480000); b.u32(0x00480000); b.u32(0);
               b.u16(1); b.zeros(32);
               b.u16(0x18); b.u16(0xffff);
-              BoxW avcC(b, "avcC");
+              BoxW avcC(b, d->codec == 1 ? "hvcC" : "avcC");
               b.bytes(d->avcc.data(), d->avcc.size());
               avcC.close();
               avc1.close();

@@ -182,3 +182,24 @@ def test_demux_parity_hevc_synthetic(built_lib):
                 pkt, _, kf = d.packet(i)
                 assert pkt == oracle_pkts[i], f"hevc packet {i} mismatch"
             assert kf is not None
+
+
+def test_remux_hevc_clip_roundtrip(built_lib):
+    """HEVC remux: span [start,end) from a sync sample -> standalone
+    hvc1 mp4 that re-demuxes with matching PTS and packets."""
+    n, delta, ts = 24, 512, 12288
+    sizes = [int(50 + i) for i in range(n)]
+    data = mp4_write.write_mp4(
+        sizes, stts=[(n, delta)], ctts=None, timescale=ts,
+        sync_samples=[1, 13], codec="hevc",
+    )
+    with hotpath.Demuxer(data) as d:
+        clip = d.remux_clip(0.0, 13 * 512 / ts)  # frames 0..12
+    with hotpath.Demuxer(clip) as d2:
+        info = d2.probe()
+        assert info.codec == 1
+        assert info.num_samples == 13
+        pkt, _, kf = d2.packet(0)
+        assert kf
+    trk = mp4_demux.parse_mp4(clip)[0]
+    assert trk.codec in ("hvc1",) and trk.hvcc
