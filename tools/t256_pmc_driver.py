@@ -9,7 +9,9 @@ from cosmos_curate_amd import hotpath  # noqa: E402
 
 lib = hotpath.require_gpu()
 stream = torch.cuda.current_stream().cuda_stream
-for (M, N, K) in [(67200, 2304, 768), (8192, 8192, 8192)]:
+import os
+shapes = {'qkv': [(67200, 2304, 768)], 'fc1': [(67200, 3072, 768)], 'sq': [(8192, 8192, 8192)], 'both': [(67200, 2304, 768), (8192, 8192, 8192)]}
+for (M, N, K) in shapes[os.environ.get('CC_PMC_SHAPE', 'both')]:
     torch.manual_seed(1)
     a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
     b = (torch.randn(N, K) * 0.3).to(torch.bfloat16).cuda()
