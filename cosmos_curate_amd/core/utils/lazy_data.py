@@ -49,3 +49,28 @@ class LazyData(Generic[T]):
 
     def __bool__(self) -> bool:
         return self.value is not None
+
+
+def prefetch(refs) -> None:
+    """Non-blocking fetch hint (ref_resolver.py:70-105 surface).
+
+    The reference tells Ray to pull ObjectRefs into the local Plasma
+    store.  The rebuild's payloads are in-process (LazyData.value), so
+    there is nothing to fetch — the function exists so stage code
+    written against the reference API runs unchanged, and so a future
+    distributed object store can slot in behind the same call.
+    """
+    _ = [r for r in refs if r is not None]
+
+
+def resolve_as_ready(items):
+    """Yield (key, value|None) pairs 1:1 with the input
+    (ref_resolver.py:108-170 surface).
+
+    The reference yields in ray.wait completion order, already-local
+    values first; with in-process payloads everything is already local,
+    so input order IS completion order.  Callers keep the reference's
+    memory discipline: process and release each item before advancing.
+    """
+    for key, ld in items:
+        yield key, (ld.resolve() if ld is not None else None)

@@ -189,3 +189,19 @@ def test_multicam_fixed_stride_shares_spans_and_uuids():
     assert [c.span for c in cams[0].clips] == [c.span for c in cams[1].clips]
     assert [c.uuid for c in cams[0].clips] == [c.uuid for c in cams[1].clips]
     assert_time_alignment(out)
+
+
+def test_lazy_data_prefetch_resolve_as_ready():
+    """ref_resolver surface (prefetch/resolve_as_ready): 1:1 mapping,
+    empty items yield (key, None)."""
+    from cosmos_curate_amd.core.utils.lazy_data import (
+        LazyData,
+        prefetch,
+        resolve_as_ready,
+    )
+
+    a = LazyData(value=b"abc", nbytes=3)
+    b = LazyData()
+    prefetch([a, b])  # no-op, must not raise
+    out = list(resolve_as_ready([("a", a), ("b", b), ("c", None)]))
+    assert out == [("a", b"abc"), ("b", None), ("c", None)]
