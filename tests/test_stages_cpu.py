@@ -205,3 +205,48 @@ def test_lazy_data_prefetch_resolve_as_ready():
     prefetch([a, b])  # no-op, must not raise
     out = list(resolve_as_ready([("a", a), ("b", b), ("c", None)]))
     assert out == [("a", b"abc"), ("b", None), ("c", None)]
+
+
+def test_timestamps_integration_real_mp4():
+    """Mirror of the reference test_timestamps_integration.py sanity
+    battery: populate_metadata/populate_timestamps on a real MP4 ->
+    monotonic finite f32 within duration; major-size delta ==
+    timestamps.nbytes; downloader -> fixed-stride on the 30 s sample
+    yields 3 clips.  Uses the reference's own fixture when present (dev
+    container), else the committed synthetic fixture."""
+    import pathlib as pl
+
+    import numpy as np
+
+    from cosmos_curate_amd.pipelines.video.utils.data_model import Video
+
+    ref = pl.Path(
+        "/root/reference/tests/cosmos_curate/pipelines/video/data/test_video_30s.mp4"
+    )
+    golden = pl.Path(__file__).parent / "golden" / "synth_bframes.mp4"
+    src = ref if ref.exists() else golden
+    v = Video(input_video=src, encoded_data=np.frombuffer(src.read_bytes(), dtype=np.uint8))
+    v.populate_metadata()
+    size_before = v.get_major_size()
+    v.populate_timestamps()
+    ts = v.timestamps
+    assert ts is not None and len(ts) > 0 and ts.dtype == np.float32
+    assert np.all(ts >= 0.0) and np.all(np.isfinite(ts))
+    assert np.all(np.diff(ts) >= 0)
+    assert v.metadata.duration is not None
+    assert float(ts[-1]) <= v.metadata.duration + 0.1
+    assert v.get_major_size() == size_before + ts.nbytes
+
+    if src is ref:
+        # downloader -> fixed-stride chain: 30 s -> 3 clips (reference
+        # test_downloader_then_fixed_stride_real_mp4...)
+        from cosmos_curate_amd.pipelines.video.read_write.download_stages import (
+            VideoDownloader,
+        )
+
+        v2 = Video(input_video=ref)
+        t = SplitPipeTask(videos=[v2])
+        VideoDownloader(input_path=str(ref.parent)).process_data([t])
+        assert "timestamps" not in v2.errors and v2.timestamps is not None
+        out = run_pipeline([t], [FixedStrideExtractorStage()], runner=SequentialRunner())
+        assert len(out[0].video.clips) == 3
