@@ -198,6 +198,10 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
           const long row = crow_base + m * FRAG + r;
           float v = acc[m][n][r] + bval;
           if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+          if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+            float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+            v = 0.5f * v * (1.0f + t);
+          }
           if constexpr (HAS_RES) v += rv[r];
           if (c_is_bf16)
             ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -222,6 +226,10 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
         if (row >= M) continue;
         float v = acc[m][n][r] + bval;
         if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));  // quick-gelu
+        if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+          float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+          v = 0.5f * v * (1.0f + t);
+        }
         if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -322,6 +330,10 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
           const long row = row0 + m * 32 + (reg & 3) + 8 * (reg >> 2);
           float v = acc[m][n][reg] + bval;
           if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+          if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+            float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+            v = 0.5f * v * (1.0f + t);
+          }
           if constexpr (HAS_RES) v += rv[reg];
           if (c_is_bf16)
             ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -346,6 +358,10 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
         if (row >= M) continue;
         float v = acc[m][n][reg] + bval;
         if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+        if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+          float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+          v = 0.5f * v * (1.0f + t);
+        }
         if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -539,6 +555,10 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
           const long row = crow_base + m * FRAG + r;
           float v = acc[m][n][r] + bval;
           if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+          if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+            float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+            v = 0.5f * v * (1.0f + t);
+          }
           if constexpr (HAS_RES) v += rv[r];
           if (c_is_bf16)
             ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -563,6 +583,10 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
         if (row >= M) continue;
         float v = acc[m][n][r] + bval;
         if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+        if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
+          float t = tanhf(0.7978845608028654f * (v + 0.044715f * v * v * v));
+          v = 0.5f * v * (1.0f + t);
+        }
         if constexpr (HAS_RES) v += (float)residual[row * N + col];
         if (c_is_bf16)
           ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
@@ -643,19 +667,18 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
                      (const __bf16*)A, (const __bf16*)B, C, bias,             \
                      (const __bf16*)residual, (long)M, (long)N, (long)K,      \
                      c_dtype == 1 ? 1 : 0, nbx, nwg, remap)
+#define CC_DISPATCH_ACT(KER, A_)                                              \
+  do {                                                                        \
+    if (hb && hr) CC_LAUNCH_GEMM(KER, A_, true, true);                        \
+    else if (hb) CC_LAUNCH_GEMM(KER, A_, true, false);                        \
+    else if (hr) CC_LAUNCH_GEMM(KER, A_, false, true);                        \
+    else CC_LAUNCH_GEMM(KER, A_, false, false);                               \
+  } while (0)
 #define CC_DISPATCH(KER)                                                      \
   do {                                                                        \
-    if (act == 1) {                                                           \
-      if (hb && hr) CC_LAUNCH_GEMM(KER, 1, true, true);                       \
-      else if (hb) CC_LAUNCH_GEMM(KER, 1, true, false);                       \
-      else if (hr) CC_LAUNCH_GEMM(KER, 1, false, true);                       \
-      else CC_LAUNCH_GEMM(KER, 1, false, false);                              \
-    } else {                                                                  \
-      if (hb && hr) CC_LAUNCH_GEMM(KER, 0, true, true);                       \
-      else if (hb) CC_LAUNCH_GEMM(KER, 0, true, false);                       \
-      else if (hr) CC_LAUNCH_GEMM(KER, 0, false, true);                       \
-      else CC_LAUNCH_GEMM(KER, 0, false, false);                              \
-    }                                                                         \
+    if (act == 1) CC_DISPATCH_ACT(KER, 1);                                    \
+    else if (act == 2) CC_DISPATCH_ACT(KER, 2);  /* tanh-gelu (SigLIP) */     \
+    else CC_DISPATCH_ACT(KER, 0);                                             \
   } while (0)
   if (wide)
     CC_DISPATCH(k_gemm_bf16_w32);
@@ -664,6 +687,7 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   else
     CC_DISPATCH(k_gemm_bf16);
 #undef CC_DISPATCH
+#undef CC_DISPATCH_ACT
 #undef CC_LAUNCH_GEMM
   hipError_t e = hipGetLastError();
   if (timed) cc::timed_end("gemm_bf16", stream, ev0, ev1);
