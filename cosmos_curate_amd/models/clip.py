@@ -30,6 +30,8 @@ _CLIP_MODEL_ID = "openai/clip-vit-base-patch32"
 
 CLIP_MEAN = (0.48145466, 0.4578275, 0.40821073)
 CLIP_STD = (0.26862954, 0.26130258, 0.27577711)
+SIGLIP_MEAN = (0.5, 0.5, 0.5)  # SiglipImageProcessor defaults
+SIGLIP_STD = (0.5, 0.5, 0.5)
 
 
 class _CLIPImageEmbeddings(torch.nn.Module):
@@ -42,10 +44,19 @@ class _CLIPImageEmbeddings(torch.nn.Module):
         hotpath.require_gpu()  # fail loudly before any lazy surprises
         self.device = torch.device("cuda")
         cfg = cw.CONFIGS[variant]
-        sd = make_clip_vit_b32_weights() if variant == "vit_b32" else cw.make_clip_vit_weights(cfg)
-        self.tower = ClipVisionTowerAMD(sd, cfg).to(self.device)
-        self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
-        self._std_arr = np.array(CLIP_STD, dtype=np.float32)
+        if variant.startswith("siglip"):
+            from cosmos_curate_amd.models.siglip_vit import SiglipVisionTowerAMD
+
+            self.tower = SiglipVisionTowerAMD(
+                cw.make_siglip_weights(cfg), cfg).to(self.device)
+            self._mean_arr = np.array(SIGLIP_MEAN, dtype=np.float32)
+            self._std_arr = np.array(SIGLIP_STD, dtype=np.float32)
+        else:
+            sd = (make_clip_vit_b32_weights() if variant == "vit_b32"
+                  else cw.make_clip_vit_weights(cfg))
+            self.tower = ClipVisionTowerAMD(sd, cfg).to(self.device)
+            self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
+            self._std_arr = np.array(CLIP_STD, dtype=np.float32)
 
     def preprocess_u8(self, frames_dev_u8: torch.Tensor) -> torch.Tensor:
         """(N,224,224,3) u8 on device -> (N,3,224,224) bf16 normalized."""
@@ -129,6 +140,8 @@ class CLIPImageEmbeddings(ModelInterface):
     def model_id_names(self) -> list[str]:
         if self._variant == "vit_l14":
             return ["openai/clip-vit-large-patch14"]  # reference clip.py:33
+        if self._variant.startswith("siglip"):
+            return ["google/siglip-large-patch16-256"]  # BASELINE config #3
         return [_CLIP_MODEL_ID]
 
     def setup(self) -> None:

@@ -34,10 +34,12 @@ class VitConfig:
     patch: int
     proj: int
     image: int = 224
+    has_cls: bool = True       # SigLIP towers have no class token
+    act: str = "quick_gelu"    # "quick_gelu" (CLIP) | "gelu_tanh" (SigLIP)
 
     @property
     def num_pos(self) -> int:
-        return (self.image // self.patch) ** 2 + 1
+        return (self.image // self.patch) ** 2 + (1 if self.has_cls else 0)
 
 
 # ViT-B/32 = transformers CLIPVisionConfig defaults (config #1/#2 embedder)
@@ -49,7 +51,58 @@ VIT_B32 = VitConfig("vit_b32", hidden=768, layers=12, heads=12,
 VIT_L14 = VitConfig("vit_l14", hidden=1024, layers=24, heads=16,
                     intermediate=4096, patch=14, proj=768)
 
-CONFIGS = {"vit_b32": VIT_B32, "vit_l14": VIT_L14}
+# SigLIP-large geometry (google/siglip-large-patch16-256 class): no CLS
+# token, tanh-gelu MLPs, attention-pooling MAP head — the SigLIP-L-class
+# embedder BASELINE config #3 names.  proj == hidden (the MAP head's
+# pooled vector is the embedding; no visual_projection).
+SIGLIP_L16_256 = VitConfig("siglip_l16_256", hidden=1024, layers=24, heads=16,
+                           intermediate=4096, patch=16, proj=1024, image=256,
+                           has_cls=False, act="gelu_tanh")
+
+CONFIGS = {"vit_b32": VIT_B32, "vit_l14": VIT_L14,
+           "siglip_l16_256": SIGLIP_L16_256}
+
+
+def make_siglip_weights(cfg: VitConfig = SIGLIP_L16_256) -> dict[str, torch.Tensor]:
+    """State dict for transformers SiglipVisionModel of this geometry
+    (key names exactly as SiglipVisionModel.named_parameters())."""
+    sd: dict[str, torch.Tensor] = {}
+    tag = cfg.name + ":"
+    sd["embeddings.patch_embedding.weight"] = _randn(
+        tag + "patch", cfg.hidden, 3, cfg.patch, cfg.patch)
+    sd["embeddings.patch_embedding.bias"] = _randn(tag + "patch.b", cfg.hidden)
+    sd["embeddings.position_embedding.weight"] = _randn(
+        tag + "pos", cfg.num_pos, cfg.hidden)
+    for i in range(cfg.layers):
+        q = f"encoder.layers.{i}."
+        for proj in ["q_proj", "k_proj", "v_proj", "out_proj"]:
+            sd[q + f"self_attn.{proj}.weight"] = _randn(
+                f"{tag}l{i}.{proj}.w", cfg.hidden, cfg.hidden)
+            sd[q + f"self_attn.{proj}.bias"] = _randn(f"{tag}l{i}.{proj}.b", cfg.hidden)
+        sd[q + "layer_norm1.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm1.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "layer_norm2.weight"] = torch.ones(cfg.hidden)
+        sd[q + "layer_norm2.bias"] = torch.zeros(cfg.hidden)
+        sd[q + "mlp.fc1.weight"] = _randn(f"{tag}l{i}.fc1.w", cfg.intermediate, cfg.hidden)
+        sd[q + "mlp.fc1.bias"] = _randn(f"{tag}l{i}.fc1.b", cfg.intermediate)
+        sd[q + "mlp.fc2.weight"] = _randn(f"{tag}l{i}.fc2.w", cfg.hidden, cfg.intermediate)
+        sd[q + "mlp.fc2.bias"] = _randn(f"{tag}l{i}.fc2.b", cfg.hidden)
+    sd["post_layernorm.weight"] = torch.ones(cfg.hidden)
+    sd["post_layernorm.bias"] = torch.zeros(cfg.hidden)
+    sd["head.probe"] = _randn(tag + "probe", 1, 1, cfg.hidden)
+    sd["head.attention.in_proj_weight"] = _randn(
+        tag + "head.inproj.w", 3 * cfg.hidden, cfg.hidden)
+    sd["head.attention.in_proj_bias"] = _randn(tag + "head.inproj.b", 3 * cfg.hidden)
+    sd["head.attention.out_proj.weight"] = _randn(
+        tag + "head.outproj.w", cfg.hidden, cfg.hidden)
+    sd["head.attention.out_proj.bias"] = _randn(tag + "head.outproj.b", cfg.hidden)
+    sd["head.layernorm.weight"] = torch.ones(cfg.hidden)
+    sd["head.layernorm.bias"] = torch.zeros(cfg.hidden)
+    sd["head.mlp.fc1.weight"] = _randn(tag + "head.fc1.w", cfg.intermediate, cfg.hidden)
+    sd["head.mlp.fc1.bias"] = _randn(tag + "head.fc1.b", cfg.intermediate)
+    sd["head.mlp.fc2.weight"] = _randn(tag + "head.fc2.w", cfg.hidden, cfg.intermediate)
+    sd["head.mlp.fc2.bias"] = _randn(tag + "head.fc2.b", cfg.hidden)
+    return sd
 
 # module-level aliases for the flagship config (bench flop accounting)
 HIDDEN, LAYERS, HEADS = VIT_B32.hidden, VIT_B32.layers, VIT_B32.heads

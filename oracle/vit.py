@@ -51,3 +51,35 @@ def embed_frames_fp32(
     out = model(pixel_values=pixel_values.float()).image_embeds
     out = out / torch.linalg.vector_norm(out, dim=-1, keepdim=True)
     return out.cpu().numpy().astype(np.float32)
+
+
+def build_reference_siglip_vision(state_dict: dict[str, torch.Tensor],
+                                  cfg_overrides: dict | None = None):
+    """Build a transformers SiglipVisionModel from a state dict made by
+    clip_weights.make_siglip_weights (key names are exactly
+    SiglipVisionModel.named_parameters()).  Default geometry =
+    google/siglip-large-patch16-256."""
+    from transformers import SiglipVisionConfig, SiglipVisionModel
+
+    kw = dict(hidden_size=1024, intermediate_size=4096, num_hidden_layers=24,
+              num_attention_heads=16, image_size=256, patch_size=16)
+    if cfg_overrides:
+        kw.update(cfg_overrides)
+    model = SiglipVisionModel(SiglipVisionConfig(**kw))
+    missing, unexpected = model.load_state_dict(state_dict, strict=False)
+    missing = [m for m in missing if "position_ids" not in m]
+    assert not missing and not unexpected, (missing, unexpected)
+    return model.float().eval()
+
+
+@torch.no_grad()
+def siglip_embed_frames_fp32(
+    model, pixel_values: npt.NDArray[np.float32] | torch.Tensor
+) -> npt.NDArray[np.float32]:
+    """(N,3,image,image) f32 SigLIP-normalized -> L2-normalized
+    MAP-pooled fp32 embeddings (pooler_output / ||.||)."""
+    if isinstance(pixel_values, np.ndarray):
+        pixel_values = torch.from_numpy(pixel_values)
+    out = model(pixel_values=pixel_values.float()).pooler_output
+    out = out / torch.linalg.vector_norm(out, dim=-1, keepdim=True)
+    return out.cpu().numpy().astype(np.float32)

@@ -358,3 +358,37 @@ def test_preprocess_patches_matches_torch_chain(lib):
     e1 = m.tower(patches=fused, n=5)
     e2 = m.tower(pixels)
     assert torch.equal(e1, e2)
+
+
+def test_siglip_gpu_vs_oracle_cosine(lib):
+    """SigLIP-L16-256 tower (BASELINE config #3 class) on the MFMA path
+    vs transformers fp32 SiglipVisionModel: cosine >= 0.999."""
+    from cosmos_curate_amd.models import clip_weights as cw
+    from cosmos_curate_amd.models.siglip_vit import SiglipVisionTowerAMD
+    from oracle.vit import build_reference_siglip_vision, siglip_embed_frames_fp32
+
+    cfg = cw.SIGLIP_L16_256
+    sd = cw.make_siglip_weights(cfg)
+    tower = SiglipVisionTowerAMD(sd, cfg).cuda()
+    torch.manual_seed(6)
+    pix = torch.rand(3, 3, 256, 256) * 2 - 1  # SigLIP-normalized range
+    got = tower(pix.cuda().to(torch.bfloat16)).cpu().numpy()
+    ref = build_reference_siglip_vision(sd)
+    want = siglip_embed_frames_fp32(ref, pix)
+    cos = (got * want).sum(axis=1)
+    assert cos.min() > 0.999, cos
+
+
+def test_siglip_model_interface_u8_path(lib):
+    """CLIPImageEmbeddings(variant='siglip_l16_256') over u8 frames: the
+    fused patches path and the pixels path agree bit-exactly."""
+    from cosmos_curate_amd.models.clip import _CLIPImageEmbeddings
+
+    m = _CLIPImageEmbeddings("siglip_l16_256")
+    torch.manual_seed(8)
+    frames = torch.randint(0, 256, (2, 256, 256, 3), dtype=torch.uint8).cuda()
+    e1 = m(frames)
+    pixels = m.preprocess_u8(frames)
+    e2 = m.tower(pixels)
+    assert torch.equal(e1, e2)
+    assert e1.shape == (2, 1024)
