@@ -41,7 +41,7 @@ def gemm_flops_per_frame(variant: str = "vit_b32") -> float:
 
     cfg = cw.CONFIGS[variant]
     g = cfg.image // cfg.patch
-    patches, tokens = g * g, g * g + 1
+    patches, tokens = g * g, cfg.num_pos  # num_pos handles no-CLS towers
     patch_k = (3 * cfg.patch * cfg.patch + 63) // 64 * 64  # kernel K padding
     per_layer = (
         2 * tokens * cfg.hidden * (3 * cfg.hidden)  # fused qkv
@@ -124,9 +124,11 @@ def main() -> None:
         "half's GEMM chain with the other's small kernels)",
     )
     ap.add_argument(
-        "--model", default="vit_b32", choices=["vit_b32", "vit_l14"],
-        help="embedder tower: vit_b32 (flagship, configs #1/#2) or vit_l14 "
-        "(the reference's CLIP model / config #3 class)",
+        "--model", default="vit_b32",
+        choices=["vit_b32", "vit_l14", "siglip_l16_256"],
+        help="embedder tower: vit_b32 (flagship, configs #1/#2), vit_l14 "
+        "(the reference's CLIP model) or siglip_l16_256 (config #3's "
+        "SigLIP-L class)",
     )
     args = ap.parse_args()
 
@@ -147,6 +149,8 @@ def main() -> None:
 
     lib = hotpath.require_gpu()
     model = _CLIPImageEmbeddings(args.model)
+    global RES
+    RES = model.tower.cfg.image  # 224 (CLIP) / 256 (SigLIP)
 
     B = args.clips
     F = B * FRAMES_PER_CLIP
