@@ -78,13 +78,14 @@ def run_cpu_baseline(n_clips: int = 8) -> dict:
 
     ref = oracle_vit.build_reference_clip_vision(make_clip_vit_b32_weights())
     y, uv = make_nv12_batch(FRAMES_PER_CLIP, seed=0)
+    res = 224  # the baseline is always the flagship B/32 oracle
     t0 = time.perf_counter()
     for _ in range(n_clips):
         rgb = np.stack(
             [
                 ocolor.resize_bilinear_u8(
                     ocolor.nv12_to_rgb(y[j], uv[j].reshape(SRC_H // 2, SRC_W // 2, 2)),
-                    RES, RES,
+                    res, res,
                 )
                 for j in range(FRAMES_PER_CLIP)
             ]
