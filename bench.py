@@ -272,7 +272,7 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "workload": "split_pipeline 1xMI355X: HIP NV12->RGB/resize + "
-                f"CLIP-{'ViT-B/32' if args.model == 'vit_b32' else 'ViT-L/14'} MFMA bf16 "
+                f"{ {'vit_b32': 'CLIP-ViT-B/32', 'vit_l14': 'CLIP-ViT-L/14', 'siglip_l16_256': 'SigLIP-L/16-256'}[args.model] } MFMA bf16 "
                 "(decode seam excluded: no librocdecode in image)",
                 "model": args.model,
                 "clips_per_step": B,
