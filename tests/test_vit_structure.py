@@ -92,3 +92,34 @@ def test_tower_matches_transformers_fp32(weights, monkeypatch):
     assert np.all(cos >= 0.999), f"cosine too low: {cos}"
     # unit norm
     np.testing.assert_allclose(np.linalg.norm(got, axis=1), 1.0, atol=1e-3)
+
+
+def test_siglip_weights_load_into_transformers():
+    """make_siglip_weights keys match SiglipVisionModel exactly and the
+    fp32 oracle produces unit-norm pooled embeddings (tiny geometry)."""
+    import torch
+
+    from cosmos_curate_amd.models.clip_weights import VitConfig, make_siglip_weights
+    from oracle.vit import build_reference_siglip_vision, siglip_embed_frames_fp32
+
+    tiny = VitConfig("siglip_l16_256", hidden=128, layers=2, heads=2,
+                     intermediate=256, patch=16, proj=128, image=32,
+                     has_cls=False, act="gelu_tanh")
+    sd = make_siglip_weights(tiny)
+    ref = build_reference_siglip_vision(
+        sd, dict(hidden_size=128, intermediate_size=256, num_hidden_layers=2,
+                 num_attention_heads=2, image_size=32, patch_size=16))
+    e = siglip_embed_frames_fp32(ref, torch.randn(2, 3, 32, 32))
+    assert e.shape == (2, 128)
+    import numpy as np
+
+    np.testing.assert_allclose((e ** 2).sum(axis=1), 1.0, rtol=1e-5)
+
+
+def test_siglip_config_geometry():
+    from cosmos_curate_amd.models.clip_weights import SIGLIP_L16_256
+
+    cfg = SIGLIP_L16_256
+    assert cfg.num_pos == 256  # no CLS token
+    assert cfg.hidden // cfg.heads == 64  # attn_mid head-dim contract
+    assert (3 * cfg.patch * cfg.patch) % 64 == 0
