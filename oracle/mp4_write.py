@@ -32,6 +32,8 @@ def write_mp4(
     height: int = 64,
     payload_byte: int = 0xAB,
     codec: str = "h264",
+    samples_per_chunk: int | None = None,
+    use_co64: bool = False,
 ) -> bytes:
     """Build an MP4 with one avc1 (or hvc1) video track and fake samples."""
     n = len(sample_sizes)
@@ -92,11 +94,24 @@ def write_mp4(
         b"stsz", 0, 0,
         struct.pack(">II", 0, n) + b"".join(struct.pack(">I", s) for s in sample_sizes),
     )
-    stsc = _full(b"stsc", 0, 0, struct.pack(">I", 1) + struct.pack(">III", 1, n, 1))
-    stco_placeholder = _full(b"stco", 0, 0, struct.pack(">I", 1) + struct.pack(">I", 0))
+    spc = samples_per_chunk if samples_per_chunk else n
+    n_chunks = (n + spc - 1) // spc
+    stsc = _full(b"stsc", 0, 0, struct.pack(">I", 1) + struct.pack(">III", 1, spc, 1))
+    co_type = b"co64" if use_co64 else b"stco"
+    co_fmt = ">Q" if use_co64 else ">I"
+    stco_placeholder = _full(
+        co_type, 0, 0,
+        struct.pack(">I", n_chunks) + struct.pack(co_fmt, 0) * n_chunks)
 
     def build_moov(chunk_off: int) -> bytes:
-        stco = _full(b"stco", 0, 0, struct.pack(">I", 1) + struct.pack(">I", chunk_off))
+        # chunk c starts at the sum of sample sizes before it (contiguous mdat)
+        offs, acc = [], chunk_off
+        for c in range(n_chunks):
+            offs.append(acc)
+            acc += sum(sample_sizes[c * spc:(c + 1) * spc])
+        stco = _full(co_type, 0, 0,
+                     struct.pack(">I", n_chunks)
+                     + b"".join(struct.pack(co_fmt, o) for o in offs))
         stbl = _box(b"stbl", stsd + stts_b + ctts_b + stss_b + stsz + stsc + stco)
         total_dur = sum(c * d for c, d in stts)
         mdhd = _full(

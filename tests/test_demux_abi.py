@@ -203,3 +203,24 @@ def test_remux_hevc_clip_roundtrip(built_lib):
         assert kf
     trk = mp4_demux.parse_mp4(clip)[0]
     assert trk.codec in ("hvc1",) and trk.hvcc
+
+
+def test_demux_parity_multichunk_co64(built_lib):
+    """Multi-chunk stsc layouts and 64-bit chunk offsets (co64): the
+    C++ demuxer's per-sample byte ranges == the oracle's."""
+    rng = np.random.default_rng(0xC064)
+    for spc, co64 in [(1, False), (3, False), (5, True), (2, True)]:
+        n = int(rng.integers(6, 30))
+        sizes = [int(rng.integers(8, 120)) for _ in range(n)]
+        data = mp4_write.write_mp4(
+            sizes, stts=[(n, 512)], ctts=None, timescale=12288,
+            sync_samples=[1], samples_per_chunk=spc, use_co64=co64,
+        )
+        trk = mp4_demux.parse_mp4(data)[0]
+        oracle_pkts = mp4_demux.annexb_packets(data, trk)
+        oracle_ts = mp4_demux.get_video_timestamps(data)
+        with hotpath.Demuxer(data) as d:
+            np.testing.assert_array_equal(d.timestamps(), oracle_ts)
+            for i in [0, n // 2, n - 1]:
+                pkt, _, _ = d.packet(i)
+                assert pkt == oracle_pkts[i], f"spc={spc} co64={co64} pkt {i}"
