@@ -333,6 +333,179 @@ __global__ __launch_bounds__(256, 2) void k_attn_mid(
   }
 }
 
+
+// ---- flash variant (seq > 288): K/V streamed in 64-row tiles with
+// online softmax ----
+// One workgroup per (frame, head, q-tile of 64 rows); Q and K fragments
+// load straight from global (16B-contiguous in the fused-QKV layout,
+// same trick as k_attn_mid); per KV-tile, V^T and the probability tile
+// stage through a small LDS image (~19 KB -> high occupancy).  f32
+// running max/sum per row, O accumulators rescaled flash-style,
+// probabilities rounded to bf16 before PV (same numerics contract as
+// the resident kernels).
+constexpr int FKV = 64;        // kv tile rows
+constexpr int LDF = FKV + 8;   // LDS s-stride
+
+__global__ __launch_bounds__(256, 2) void k_attn_flash(
+    const __bf16* __restrict__ qkv, __bf16* __restrict__ out, long n_frames,
+    int seq, int heads, int hidden, float scale) {
+  const int n_qtiles = (seq + 63) / 64;
+  const long fhq = blockIdx.x;
+  const int qt = (int)(fhq % n_qtiles);
+  const long fh = fhq / n_qtiles;
+  const long frame = fh / heads;
+  const int head = fh % heads;
+  if (frame >= n_frames) return;
+
+  __shared__ __bf16 lds[2 * 64 * LDF];
+  __bf16* Vt = lds;              // [64 d][LDF s]
+  __bf16* P = lds + 64 * LDF;    // [64 q][LDF s]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const long qkv_base = (frame * (long)seq * 3) * hidden + (long)head * HD;
+  const int q0 = qt * 64;
+
+  // per-lane row state (rows rbase..rbase+3 of this wave's 16)
+  const int rbase_l = 4 * (lane >> 4);
+  float m_run[4], l_run[4];
+  f32x4 oacc[4] = {};
+#pragma unroll
+  for (int r = 0; r < 4; r++) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.0f;
+  }
+
+  const int n_kv = (seq + FKV - 1) / FKV;
+  for (int kt = 0; kt < n_kv; kt++) {
+    const int k0 = kt * FKV;
+    // ---- V^T tile into LDS (thread t -> row t>>2, d (t&3)*16)
+    __syncthreads();  // P/Vt free from the previous tile
+    {
+      const int row = tid >> 2;
+      const int d0 = (tid & 3) * 16;
+      bf16x8 z = {};
+      bf16x8 v0 = z, v1 = z;
+      if (k0 + row < seq) {
+        const long base = qkv_base + (long)(k0 + row) * 3 * hidden + 2 * hidden;
+        v0 = *(const bf16x8*)(qkv + base + d0);
+        v1 = *(const bf16x8*)(qkv + base + d0 + 8);
+      }
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        Vt[(d0 + e) * LDF + row] = v0[e];
+        Vt[(d0 + 8 + e) * LDF + row] = v1[e];
+      }
+    }
+
+    // ---- S tile = Q K^T (wave rows, 4 col frags), direct global Q/K
+    f32x4 acc[4] = {};
+    {
+      int qrow = q0 + 16 * wid + (lane & 15);
+      qrow = qrow < seq ? qrow : seq - 1;
+      const int k0e = 8 * (lane >> 4);
+#pragma unroll
+      for (int kk = 0; kk < HD; kk += 32) {
+        bf16x8 qf =
+            *(const bf16x8*)(qkv + qkv_base + (long)qrow * 3 * hidden + kk + k0e);
+#pragma unroll
+        for (int n = 0; n < 4; n++) {
+          int krow = k0 + n * 16 + (lane & 15);
+          krow = krow < seq ? krow : seq - 1;
+          bf16x8 kf = *(const bf16x8*)(qkv + qkv_base + (long)krow * 3 * hidden +
+                                       hidden + kk + k0e);
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, kf, acc[n], 0, 0, 0);
+        }
+      }
+    }
+
+    // ---- online softmax update (per row)
+    float alpha[4];
+    {
+      const int colr = lane & 15;
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        float mx = m_run[reg];
+        float sv[4];
+#pragma unroll
+        for (int nn = 0; nn < 4; nn++) {
+          float sc = acc[nn][reg] * scale;
+          if (k0 + nn * 16 + colr >= seq) sc = -1e30f;
+          sv[nn] = sc;
+          mx = fmaxf(mx, sc);
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off));
+        alpha[reg] = __expf(m_run[reg] - mx);
+        float sum = 0.0f;
+#pragma unroll
+        for (int nn = 0; nn < 4; nn++) {
+          float e = __expf(sv[nn] - mx);
+          if (k0 + nn * 16 + colr >= seq) e = 0.0f;
+          sv[nn] = e;
+          sum += e;
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
+        l_run[reg] = l_run[reg] * alpha[reg] + sum;
+        m_run[reg] = mx;
+#pragma unroll
+        for (int nn = 0; nn < 4; nn++) acc[nn][reg] = sv[nn];
+      }
+    }
+    // ---- P tile to LDS (unnormalized probabilities, bf16)
+    {
+      const int colr = lane & 15;
+      const int rb = 16 * wid + rbase_l;
+#pragma unroll
+      for (int nn = 0; nn < 4; nn++)
+#pragma unroll
+        for (int reg = 0; reg < 4; reg++)
+          P[(rb + reg) * LDF + nn * 16 + colr] = (__bf16)acc[nn][reg];
+    }
+    __syncthreads();
+
+    // ---- O = alpha * O + P V
+    f32x4 pv[4] = {};
+    {
+      const int prow = 16 * wid + (lane & 15);
+      const int j0 = 8 * (lane >> 4);
+#pragma unroll
+      for (int kk = 0; kk < FKV; kk += 32) {
+        bf16x8 pf = *(const bf16x8*)(P + prow * LDF + kk + j0);
+#pragma unroll
+        for (int n = 0; n < 4; n++) {
+          bf16x8 vf =
+              *(const bf16x8*)(Vt + (n * 16 + (lane & 15)) * LDF + kk + j0);
+          pv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, pv[n], 0, 0, 0);
+        }
+      }
+    }
+#pragma unroll
+    for (int n = 0; n < 4; n++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++)
+        oacc[n][reg] = oacc[n][reg] * alpha[reg] + pv[n][reg];
+  }
+
+  // ---- store O / l
+  {
+    const int dcol = lane & 15;
+    const int rb = 16 * wid + rbase_l;
+#pragma unroll
+    for (int nn = 0; nn < 4; nn++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int row = q0 + rb + reg;
+        if (row < seq)
+          out[(frame * seq + row) * (long)hidden + head * HD + nn * 16 + dcol] =
+              (__bf16)(oacc[nn][reg] / l_run[reg]);
+      }
+  }
+}
+
 }  // namespace
 
 extern "C" int cc_attn_small(const void* qkv, void* out, int64_t n_frames,
@@ -374,5 +547,28 @@ extern "C" int cc_attn_mid(const void* qkv, void* out, int64_t n_frames,
   if (timed) cc::timed_end("attn_mid", stream, ev0, ev1);
   if (e != hipSuccess)
     return cc::set_error(CC_ERR_HIP, "attn_mid launch: %s", hipGetErrorString(e));
+  return CC_OK;
+}
+
+extern "C" int cc_attn_flash(const void* qkv, void* out, int64_t n_frames,
+                             int seq, int heads, int hidden, float scale,
+                             uint64_t stream) {
+  if (!qkv || !out || n_frames <= 0 || seq <= 0)
+    return cc::set_error(CC_ERR_INVALID, "bad attn args");
+  if (seq <= SMID || hidden != heads * HD)
+    return cc::set_error(CC_ERR_UNSUPPORTED,
+                         "cc_attn_flash is for seq > 288 with hd == 64 "
+                         "(use attn_small/attn_mid below that)");
+  const int n_qtiles = (seq + 63) / 64;
+  dim3 block(256), grid((unsigned)(n_frames * heads * (long)n_qtiles));
+  hipEvent_t ev0, ev1;
+  bool timed = cc::timed_begin(stream, &ev0, &ev1);
+  hipLaunchKernelGGL(k_attn_flash, grid, block, 0, (hipStream_t)stream,
+                     (const __bf16*)qkv, (__bf16*)out, (long)n_frames, seq,
+                     heads, hidden, scale);
+  hipError_t e = hipGetLastError();
+  if (timed) cc::timed_end("attn_flash", stream, ev0, ev1);
+  if (e != hipSuccess)
+    return cc::set_error(CC_ERR_HIP, "attn_flash launch: %s", hipGetErrorString(e));
   return CC_OK;
 }

@@ -78,6 +78,7 @@ def _declare(lib: ctypes.CDLL) -> None:
         c.c_void_p, c.c_void_p, c.c_int64, c.c_int, c.c_int, c.c_int,
         c.c_float, c.c_uint64]
     lib.cc_attn_mid.argtypes = lib.cc_attn_small.argtypes
+    lib.cc_attn_flash.argtypes = lib.cc_attn_small.argtypes
     lib.cc_layernorm_bf16.argtypes = [
         c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p, c.c_int64, c.c_int64,
         c.c_float, c.c_uint64]

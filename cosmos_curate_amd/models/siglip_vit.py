@@ -165,14 +165,16 @@ class SiglipVisionTowerAMD(torch.nn.Module):
                 getattr(self, f"w_qkv_{i}"),
                 getattr(self, f"b_qkv_{i}"),
             )
-            if qkv_flat.is_cuda and seq <= 288 and hd == 64:
+            if qkv_flat.is_cuda and hd == 64:
                 lib = hotpath.require_gpu()
                 attn = torch.empty(
                     (n * seq, cfg.hidden), dtype=torch.bfloat16,
                     device=qkv_flat.device,
                 )
                 stream = torch.cuda.current_stream(qkv_flat.device).cuda_stream
-                attn_fn = lib.cc_attn_small if seq <= 64 else lib.cc_attn_mid
+                attn_fn = (lib.cc_attn_small if seq <= 64 else
+                           lib.cc_attn_mid if seq <= 288 else
+                           lib.cc_attn_flash)
                 hotpath.check(attn_fn(
                     qkv_flat.data_ptr(), attn.data_ptr(), n, seq, self.heads,
                     cfg.hidden, ctypes.c_float(self.scale), stream,

@@ -178,6 +178,11 @@ int cc_attn_small(const void* qkv, void* out, int64_t n_frames, int seq,
 int cc_attn_mid(const void* qkv, void* out, int64_t n_frames, int seq,
                 int heads, int hidden, float scale, uint64_t stream);
 
+/* flash variant (seq > 288, head dim 64): K/V streamed in 64-row tiles
+ * with online softmax — covers SigLIP-384/so400m-class towers. */
+int cc_attn_flash(const void* qkv, void* out, int64_t n_frames, int seq,
+                  int heads, int hidden, float scale, uint64_t stream);
+
 /* ---- fused bf16 LayerNorm (replaces torch layer_norm in the ViT
  * forward; f32 stats/affine, H multiple of 256). */
 int cc_layernorm_bf16(const void* x, const void* w, const void* b, void* y,

@@ -392,3 +392,29 @@ def test_siglip_model_interface_u8_path(lib):
     e2 = m.tower(pixels)
     assert torch.equal(e1, e2)
     assert e1.shape == (2, 1024)
+
+
+@pytest.mark.parametrize("seq", [576, 700, 1024])
+def test_attn_flash_vs_sdpa(lib, seq):
+    """cc_attn_flash (online-softmax K/V streaming) == torch sdpa at
+    SigLIP-384-class and larger sequence lengths."""
+    import ctypes
+
+    torch.manual_seed(seq)
+    n, heads, hd = 2, 4, 64
+    hidden = heads * hd
+    qkv = (torch.randn(n * seq, 3 * hidden) * 0.5).to(torch.bfloat16).cuda()
+    out = torch.empty(n * seq, hidden, dtype=torch.bfloat16, device="cuda")
+    scale = hd ** -0.5
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(lib.cc_attn_flash(
+        qkv.data_ptr(), out.data_ptr(), n, seq, heads, hidden,
+        ctypes.c_float(scale), stream))
+    torch.cuda.synchronize()
+    q3 = qkv.reshape(n, seq, 3, heads, hd)
+    q = q3[:, :, 0].permute(0, 2, 1, 3).float()
+    k = q3[:, :, 1].permute(0, 2, 1, 3).float()
+    v = q3[:, :, 2].permute(0, 2, 1, 3).float()
+    want = torch.nn.functional.scaled_dot_product_attention(q, k, v, scale=scale)
+    want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
+    torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
