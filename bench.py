@@ -271,7 +271,7 @@ def main() -> None:
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "workload": "split_pipeline 1xMI355X: HIP NV12->RGB/resize + "
+                "workload": f"split_pipeline {world}xMI355X: HIP NV12->RGB/resize + "
                 f"{ {'vit_b32': 'CLIP-ViT-B/32', 'vit_l14': 'CLIP-ViT-L/14', 'siglip_l16_256': 'SigLIP-L/16-256'}[args.model] } MFMA bf16 "
                 "(decode seam excluded: no librocdecode in image)",
                 "model": args.model,
