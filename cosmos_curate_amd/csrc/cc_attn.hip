@@ -1,17 +1,21 @@
-// Fused short-sequence multi-head attention for the ViT encoder.
+// Fused multi-head attention for the ViT encoders — three ladder rungs
+// by sequence length, all reading Q/K/V straight out of the fused-QKV
+// GEMM output ([n*seq, 3*H] rows tokens) and writing O contiguous
+// [n*seq, H] (the out-projection GEMM's input layout), replacing
+// torch-rocm sdpa (~94 TF/s) + the qkv permute copies:
+//   k_attn_small  seq <= 64   (ViT-B/32's 50): everything LDS-resident,
+//                 4 WG... 2 WG/CU — see its note on the direct-global
+//                 variant that measured slower here;
+//   k_attn_mid    64 < seq <= 288 (ViT-L/14's 257, SigLIP-L16-256's
+//                 256): Q/K fragments direct from global (16B-contiguous
+//                 rows), LDS only V^T + P -> 2 WG/CU;
+//   k_attn_flash  seq > 288 (SigLIP-384-class): K/V streamed in 64-row
+//                 tiles with online softmax, O rescaled flash-style.
 //
-// Replaces torch-rocm's sdpa (attn_fwd, ~94 TF/s at the bench shape) plus
-// the qkv-slice permute copies around it: reads Q/K/V straight out of the
-// fused-QKV GEMM output (layout [n*seq, 3*H], rows tokens), computes
-// O = softmax(Q K^T / sqrt(hd)) V per (frame, head) entirely in one
-// workgroup (LDS-resident, seq <= 64, hd = 64), and writes O contiguous
-// [n*seq, H] — the exact input layout of the out-projection GEMM.
-// ViT-B/32: seq=50, heads=12, n=frames -> n*12 workgroups.
-// (ViT-L/14's seq=257 exceeds the LDS tile; it stays on torch sdpa.)
-//
-// Numerics: f32 accumulation and softmax (matches sdpa's f32 softmax on
-// bf16 inputs); probabilities round to bf16 before PV (flash-style),
-// covered by the end-to-end cosine tests.
+// Numerics (all rungs): f32 accumulation and softmax (matches sdpa's
+// f32 softmax on bf16 inputs); probabilities round to bf16 before PV,
+// covered by sdpa-parity tests at seq 50/100/257/288/576/700/1024 and
+// the end-to-end cosine tests.
 
 #include <hip/hip_runtime.h>
 
