@@ -142,3 +142,35 @@ def test_nv12_rejects_odd_dims(lib):
     rc = lib.cc_nv12_to_rgb_resize(buf.data_ptr(), buf.data_ptr(), 1, 64, 63, 64,
                                    out.data_ptr(), 32, 32, 0)
     assert rc != 0
+
+
+def test_error_paths_loud():
+    """ABI error convention: unsupported shapes return negative codes
+    with a message, never silently fall back."""
+    import torch
+
+    from cosmos_curate_amd import hotpath
+
+    lib = hotpath.require_gpu()
+    a = torch.zeros(64, 3 * 256, dtype=torch.bfloat16, device="cuda")
+    o = torch.zeros(64, 256, dtype=torch.bfloat16, device="cuda")
+    # attn_flash refuses seq <= 288
+    rc = lib.cc_attn_flash(a.data_ptr(), o.data_ptr(), 1, 64, 4, 256,
+                           ctypes.c_float(0.125), 0)
+    assert rc < 0 and b"attn_flash" in lib.cc_last_error()
+    # attn_mid refuses seq <= 64
+    rc = lib.cc_attn_mid(a.data_ptr(), o.data_ptr(), 1, 64, 4, 256,
+                         ctypes.c_float(0.125), 0)
+    assert rc < 0
+    # preprocess_patches refuses kpad not covering 3*P*P
+    frames = torch.zeros(1, 224, 224, 3, dtype=torch.uint8, device="cuda")
+    out = torch.zeros(49, 64, dtype=torch.bfloat16, device="cuda")
+    mean = (ctypes.c_float * 3)(0, 0, 0)
+    std = (ctypes.c_float * 3)(1, 1, 1)
+    rc = lib.cc_clip_preprocess_patches(frames.data_ptr(), 1, 224, 224, 32,
+                                        64, mean, std, out.data_ptr(), 0)
+    assert rc < 0
+    # layernorm refuses unsupported H
+    rc = lib.cc_layernorm_bf16(a.data_ptr(), a.data_ptr(), a.data_ptr(),
+                               o.data_ptr(), 4, 192, 1e-5, 0)
+    assert rc < 0
