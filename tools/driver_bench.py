@@ -25,8 +25,10 @@ from cosmos_curate_amd.pipelines.video.splitting_pipeline import (  # noqa: E402
 from cosmos_curate_amd.pipelines.video.utils import raw_backend  # noqa: E402
 
 
-def main(n_videos: int = 16, secs: int = 30) -> None:
-    tmp = pathlib.Path(tempfile.mkdtemp(prefix="drvbench_"))
+def main(n_videos: int = 6, secs: int = 10) -> None:
+    # raw NV12 is ~93 MB/s of video: keep the corpus small and in shm
+    base = "/dev/shm" if pathlib.Path("/dev/shm").is_dir() else None
+    tmp = pathlib.Path(tempfile.mkdtemp(prefix="drvbench_", dir=base))
     try:
         inp = tmp / "in"
         inp.mkdir()
