@@ -228,6 +228,8 @@ def _slice_raw_clip(raw: bytes, span: tuple[float, float]) -> bytes:
     last = int(math.ceil(span[1] * fps - 1e-6))  # exclusive
     first = max(0, min(first, n))
     last = max(first, min(last, n))
+    if first == 0 and last == n:
+        return raw  # span covers the whole source: zero-copy payload
     idx = np.arange(first, last, dtype=np.int32)
     ys, uvs = raw_backend.frame_planes(raw, idx)
     num = int(round(fps))
