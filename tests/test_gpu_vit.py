@@ -418,3 +418,23 @@ def test_attn_flash_vs_sdpa(lib, seq):
     want = torch.nn.functional.scaled_dot_product_attention(q, k, v, scale=scale)
     want = want.permute(0, 2, 1, 3).reshape(n * seq, hidden)
     torch.testing.assert_close(out.float(), want, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize(("M", "N", "K"), [(200, 256, 128), (300, 1600, 1536)])
+def test_gemm_tanh_gelu_epilogue(lib, M, N, K):
+    """ACT=2 (gelu_pytorch_tanh, SigLIP) epilogue on both GEMM bodies."""
+    torch.manual_seed(9)
+    a = torch.randn(M, K).to(torch.bfloat16).cuda()
+    b = torch.randn(N, K).to(torch.bfloat16).cuda()
+    bias = torch.randn(N).float().cuda()
+    res = torch.randn(M, N).to(torch.bfloat16).cuda()
+    out = torch.empty((M, N), dtype=torch.bfloat16, device="cuda")
+    stream = torch.cuda.current_stream().cuda_stream
+    hotpath.check(
+        lib.cc_gemm_bf16_ex(a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K,
+                            bias.data_ptr(), 1, 2, res.data_ptr(), stream)
+    )
+    torch.cuda.synchronize()
+    y = a.float().cpu() @ b.float().cpu().T + bias.cpu()
+    y = torch.nn.functional.gelu(y, approximate="tanh") + res.float().cpu()
+    torch.testing.assert_close(out.float().cpu(), y, rtol=2e-2, atol=8e-2)
