@@ -295,6 +295,7 @@ class ClipTranscodingStage(CuratorStage):
             return
         raw = bytes(data) if not isinstance(data, bytes) else data
         is_raw = raw_backend.is_raw_nv12(raw)
+        video.was_remuxed = not is_raw  # mp4 spans get new containers
         for clip in video.clips:
             try:
                 payload = _slice_raw_clip(raw, clip.span) if is_raw else _remux_mp4_clip(raw, clip.span)

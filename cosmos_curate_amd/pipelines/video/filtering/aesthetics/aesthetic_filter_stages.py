@@ -102,6 +102,7 @@ class AestheticFilterStage(CuratorStage):
                         else:
                             video.filtered_clips.append(clip)
                             video.clip_stats.num_filtered += 1
+                            video.clip_stats.num_filtered_by_aesthetic += 1
                     video.clips = passed
             if self._log_stats:
                 name, stats = self._timer.log_stats()

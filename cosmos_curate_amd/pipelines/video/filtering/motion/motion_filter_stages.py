@@ -175,6 +175,7 @@ class MotionFilterStage(CuratorStage):
                         if small and not self._score_only:
                             video.filtered_clips.append(clip)
                             video.clip_stats.num_filtered += 1
+                            video.clip_stats.num_filtered_by_motion += 1
                         else:
                             kept.append(clip)
                     video.clips = kept

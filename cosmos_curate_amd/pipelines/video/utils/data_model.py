@@ -123,6 +123,12 @@ class ClipStats:
     num_transcoded: int = 0
     num_with_embeddings: int = 0
     num_with_errors: int = 0
+    # per-filter counters + duration rollups (data_model.py:345-390 subset
+    # for the filters this rebuild ships)
+    num_filtered_by_aesthetic: int = 0
+    num_filtered_by_motion: int = 0
+    total_clip_duration: float = 0.0
+    max_clip_duration: float = 0.0
 
     def combine(self, other: "ClipStats") -> None:
         self.num_clips += other.num_clips
@@ -130,6 +136,10 @@ class ClipStats:
         self.num_passed += other.num_passed
         self.num_transcoded += other.num_transcoded
         self.num_with_embeddings += other.num_with_embeddings
+        self.num_filtered_by_aesthetic += other.num_filtered_by_aesthetic
+        self.num_filtered_by_motion += other.num_filtered_by_motion
+        self.total_clip_duration += other.total_clip_duration
+        self.max_clip_duration = max(self.max_clip_duration, other.max_clip_duration)
         self.num_with_errors += other.num_with_errors
 
 
@@ -149,6 +159,7 @@ class Video:
     num_clip_chunks: int = 0
     clip_chunk_index: int = 0
     clip_stats: ClipStats = dataclasses.field(default_factory=ClipStats)
+    was_remuxed: bool = False  # transcode produced new containers (:was_remuxed)
     errors: dict[str, str] = dataclasses.field(default_factory=dict)
 
     def __post_init__(self) -> None:
