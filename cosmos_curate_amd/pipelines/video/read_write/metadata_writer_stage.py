@@ -114,6 +114,10 @@ class ClipWriterStage(CuratorStage):
                         if clip.errors:
                             video.clip_stats.num_with_errors += 1
                         video.clip_stats.num_clips += 1
+                        dur = float(clip.span[1] - clip.span[0])
+                        video.clip_stats.total_clip_duration += dur
+                        video.clip_stats.max_clip_duration = max(
+                            video.clip_stats.max_clip_duration, dur)
                 self._write_embeddings_parquet(emb_rows)
             if self._log_stats:
                 name, stats = self._timer.log_stats()
