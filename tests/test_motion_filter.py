@@ -96,6 +96,7 @@ def test_motion_stage_filters_small_motion():
     assert kept.decoded_motion_data is None  # payload freed
     filt = video.filtered_clips[0]
     assert filt.motion_score_global_mean == 0.0
+    assert video.clip_stats.num_filtered_by_motion == 1  # per-filter counter
 
 
 def test_missing_motion_data_records_error():
