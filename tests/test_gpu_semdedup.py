@@ -79,3 +79,40 @@ def test_pairwise_kernel_full_size_property():
     kept = maxv <= 0.99
     assert not kept[dup_dst].any()
     assert kept[non_planted].all()
+
+
+def test_dedup_pipeline_cli(tmp_path):
+    """Standalone dedup pipeline (reference dedup_pipeline.py shape):
+    embeddings parquet in -> results parquet + summary out, planted
+    near-duplicates pruned."""
+    import json
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    import torch
+
+    from cosmos_curate_amd.pipelines.video.dedup_pipeline import cli_run_dedup
+
+    rng = np.random.default_rng(12)
+    m, d = 600, 512
+    emb = rng.normal(size=(m, d)).astype(np.float32)
+    emb[:60] = emb[60:120] + 0.001 * rng.normal(size=(60, d)).astype(np.float32)
+    ed = tmp_path / "clip_embd"
+    ed.mkdir()
+    pq.write_table(
+        pa.table({"id": [f"c{i}" for i in range(m)],
+                  "embedding": [e for e in emb]}),
+        ed / "chunk_0.parquet",
+    )
+    out = tmp_path / "dedup"
+    summary = cli_run_dedup([
+        "--input-embeddings-path", str(tmp_path),
+        "--output-dedup-path", str(out),
+        "--n-clusters", "8", "--eps", "0.05",
+    ])
+    assert summary["num_embeddings"] == m
+    assert summary["num_removed"] >= 55  # the planted near-dupes
+    t = pq.read_table(out / "dedup_results.parquet")
+    assert t.num_rows == m and set(t.column_names) == {"id", "cluster", "keep"}
+    disk = json.loads((out / "summary.json").read_text())
+    assert disk["num_kept"] == summary["num_kept"]
