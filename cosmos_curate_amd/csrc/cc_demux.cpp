@@ -57,6 +57,21 @@ inline uint32_t rd32(const uint8_t* p) {
   return ((uint32_t)p[0] << 24) | (p[1] << 16) | (p[2] << 8) | p[3];
 }
 inline int32_t rd32s(const uint8_t* p) { return (int32_t)rd32(p); }
+
+// robustness caps for corrupt containers (fuzz contract: clean error or
+// bounded parse, never a hang / OOB / memory blowup)
+constexpr size_t CC_MAX_SAMPLES = 1u << 24;
+
+struct Box;  // fwd
+template <typename BoxT>
+inline uint32_t clamp_entries(uint32_t n, const BoxT& bx, size_t hdr,
+                              size_t entry) {
+  size_t body = bx.body_end > bx.body_start + hdr
+                    ? bx.body_end - bx.body_start - hdr
+                    : 0;
+  size_t fit = body / entry;
+  return n < fit ? n : (uint32_t)fit;
+}
 inline uint16_t rd16(const uint8_t* p) { return (uint16_t)((p[0] << 8) | p[1]); }
 inline uint64_t rd64(const uint8_t* p) {
   return ((uint64_t)rd32(p) << 32) | rd32(p + 4);
@@ -83,6 +98,7 @@ struct cc_demux {
 };
 
 static bool parse_stbl(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
+  bool bad = false;  // corrupt-table flag: normalize, then reject
   std::vector<std::pair<uint32_t, uint32_t>> stsc;  // first_chunk, samples_per_chunk
   std::vector<uint64_t> chunk_offsets;
   BoxIter it{d, b, e};
@@ -94,11 +110,14 @@ static bool parse_stbl(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
       size_t pos = bx.body_start + 8;
       for (uint32_t i = 0; i < n && pos + 8 <= bx.body_end; i++) {
         uint32_t esize = rd32(d + pos);
+        // corrupt entry sizes must neither escape the box nor stall the
+        // loop (esize 0) nor let the nested iterator read past the box
+        if (esize < 16 || esize > bx.body_end - pos) break;
         char fmt[5] = {0};
         memcpy(fmt, d + pos + 4, 4);
         if (!strcmp(fmt, "avc1") || !strcmp(fmt, "avc3")) t->codec = 0;
         if (!strcmp(fmt, "hvc1") || !strcmp(fmt, "hev1")) t->codec = 1;
-        if (t->codec >= 0) {
+        if (t->codec >= 0 && esize >= 86) {
           t->width = rd16(d + pos + 32);
           t->height = rd16(d + pos + 34);
           BoxIter it2{d, pos + 86, pos + esize};
@@ -111,52 +130,70 @@ static bool parse_stbl(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
         pos += esize;
       }
     } else if (!strcmp(bx.type, "stts")) {
-      uint32_t n = rd32(p + 4);
+      // corrupt-input bounds (fuzz-found hang: a flipped count byte made
+      // the run expansion loop effectively unbounded): entry lists are
+      // clamped to what fits in the box, expansions to CC_MAX_SAMPLES.
+      uint32_t n = clamp_entries(rd32(p + 4), bx, 8, 8);
       int64_t tcur = 0;
       size_t pos = bx.body_start + 8;
       for (uint32_t i = 0; i < n; i++, pos += 8) {
         uint32_t cnt = rd32(d + pos), delta = rd32(d + pos + 4);
         for (uint32_t j = 0; j < cnt; j++) {
+          if (t->dts.size() >= CC_MAX_SAMPLES) { bad = true; break; }
           t->dts.push_back(tcur);
           tcur += delta;
         }
+        if (bad) break;
       }
     } else if (!strcmp(bx.type, "ctts")) {
-      uint32_t n = rd32(p + 4);
+      uint32_t n = clamp_entries(rd32(p + 4), bx, 8, 8);
       size_t pos = bx.body_start + 8;
       for (uint32_t i = 0; i < n; i++, pos += 8) {
         uint32_t cnt = rd32(d + pos);
         int32_t off = rd32s(d + pos + 4);
-        for (uint32_t j = 0; j < cnt; j++) t->cts.push_back(off);
+        for (uint32_t j = 0; j < cnt; j++) {
+          if (t->cts.size() >= CC_MAX_SAMPLES) { bad = true; break; }
+          t->cts.push_back(off);
+        }
+        if (bad) break;
       }
     } else if (!strcmp(bx.type, "stss")) {
-      uint32_t n = rd32(p + 4);
+      uint32_t n = clamp_entries(rd32(p + 4), bx, 8, 4);
       for (uint32_t i = 0; i < n; i++) t->sync.push_back(rd32(p + 8 + 4 * i));
     } else if (!strcmp(bx.type, "stsz")) {
       uint32_t ss = rd32(p + 4), n = rd32(p + 8);
-      if (ss)
+      if (ss) {
+        if (n > CC_MAX_SAMPLES) { bad = true; n = 0; }
         t->sizes.assign(n, ss);
-      else
+      } else {
+        n = clamp_entries(n, bx, 12, 4);
         for (uint32_t i = 0; i < n; i++) t->sizes.push_back(rd32(p + 12 + 4 * i));
+      }
     } else if (!strcmp(bx.type, "stsc")) {
-      uint32_t n = rd32(p + 4);
+      uint32_t n = clamp_entries(rd32(p + 4), bx, 8, 12);
       for (uint32_t i = 0; i < n; i++)
         stsc.push_back({rd32(p + 8 + 12 * i), rd32(p + 12 + 12 * i)});
     } else if (!strcmp(bx.type, "stco") || !strcmp(bx.type, "co64")) {
-      uint32_t n = rd32(p + 4);
       bool w = !strcmp(bx.type, "co64");
+      uint32_t n = clamp_entries(rd32(p + 4), bx, 8, w ? 8 : 4);
       for (uint32_t i = 0; i < n; i++)
         chunk_offsets.push_back(w ? rd64(p + 8 + 8 * i) : rd32(p + 8 + 4 * i));
     }
   }
   if (t->cts.empty()) t->cts.assign(t->dts.size(), 0);
-  // expand chunk map to per-sample offsets
+  // corrupt containers can leave the tables inconsistent: normalize so
+  // every per-sample array shares one bounded length (no OOB indexing
+  // anywhere downstream)
+  t->cts.resize(t->dts.size(), 0);
+  // expand chunk map to per-sample offsets (clamped: corrupt stsc
+  // first_chunk fields must not produce unbounded runs)
   if (!chunk_offsets.empty() && !stsc.empty() && !t->sizes.empty()) {
     std::vector<uint32_t> per_chunk;
     for (size_t i = 0; i < stsc.size(); i++) {
       uint32_t last = (i + 1 < stsc.size()) ? stsc[i + 1].first - 1
                                             : (uint32_t)chunk_offsets.size();
-      for (uint32_t c = stsc[i].first; c <= last; c++)
+      if (last > chunk_offsets.size()) last = (uint32_t)chunk_offsets.size();
+      for (uint32_t c = stsc[i].first; c != 0 && c <= last; c++)
         per_chunk.push_back(stsc[i].second);
     }
     size_t si = 0;
@@ -169,7 +206,15 @@ static bool parse_stbl(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
       }
     }
   }
-  return true;
+  // one consistent sample count across all per-sample tables
+  size_t ns = t->dts.size();
+  ns = std::min(ns, t->sizes.size());
+  ns = std::min(ns, t->offsets.size());
+  t->dts.resize(ns);
+  t->cts.resize(ns, 0);
+  t->sizes.resize(ns);
+  t->offsets.resize(ns);
+  return !bad;
 }
 
 static bool parse_trak(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
@@ -182,12 +227,13 @@ static bool parse_trak(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
       BoxIter it2{d, bx.body_start, bx.body_end};
       Box b2;
       while (it2.next(&b2)) {
-        if (!strcmp(b2.type, "elst")) {
+        if (!strcmp(b2.type, "elst") && b2.body_end >= b2.body_start + 8) {
           const uint8_t* q = d + b2.body_start;
           uint8_t ver = q[0];
           uint32_t n = rd32(q + 4);
           size_t pos = b2.body_start + 8;
-          for (uint32_t i = 0; i < n; i++) {
+          const size_t esz = (ver == 1) ? 20 : 12;
+          for (uint32_t i = 0; i < n && pos + esz <= b2.body_end; i++) {
             int64_t media_time;
             if (ver == 1) {
               media_time = (int64_t)rd64(d + pos + 8);
@@ -208,16 +254,19 @@ static bool parse_trak(const uint8_t* d, size_t b, size_t e, cc_demux* t) {
       Box b2;
       while (it2.next(&b2)) {
         const uint8_t* q = d + b2.body_start;
-        if (!strcmp(b2.type, "mdhd")) {
+        if (!strcmp(b2.type, "mdhd") && b2.body_end >= b2.body_start + 24) {
           uint8_t ver = q[0];
           t->timescale = rd32(q + (ver == 1 ? 20 : 12));
-        } else if (!strcmp(b2.type, "hdlr")) {
+        } else if (!strcmp(b2.type, "hdlr") &&
+                   b2.body_end >= b2.body_start + 12) {
           if (!memcmp(q + 8, "vide", 4)) is_video = true;
         } else if (!strcmp(b2.type, "minf")) {
           BoxIter it3{d, b2.body_start, b2.body_end};
           Box b3;
           while (it3.next(&b3))
-            if (!strcmp(b3.type, "stbl")) parse_stbl(d, b3.body_start, b3.body_end, t);
+            if (!strcmp(b3.type, "stbl") &&
+                !parse_stbl(d, b3.body_start, b3.body_end, t))
+              return false;  // corrupt sample tables: reject the track
         }
       }
     }
@@ -356,10 +405,11 @@ int cc_demux_packet(cc_demux* d, size_t index, const uint8_t** pkt, size_t* size
   if (!d || !pkt || !size) return cc::set_error(CC_ERR_INVALID, "null arg");
   if (index >= d->sizes.size() || index >= d->offsets.size())
     return cc::set_error(CC_ERR_INVALID, "sample index out of range");
-  const uint8_t* sample = d->data.data() + d->offsets[index];
   size_t ssize = d->sizes[index];
-  if (d->offsets[index] + ssize > d->data.size())
+  if (d->offsets[index] > d->data.size() ||
+      ssize > d->data.size() - d->offsets[index])
     return cc::set_error(CC_ERR_PARSE, "sample range outside file");
+  const uint8_t* sample = d->data.data() + d->offsets[index];
   bool is_sync = d->sync.empty() ||
                  std::binary_search(d->sync.begin(), d->sync.end(), (uint32_t)(index + 1));
   d->pkt_buf.clear();
@@ -466,6 +516,9 @@ int cc_demux_remux_clip(cc_demux* d, double start_s, double end_s,
   std::vector<std::pair<uint32_t, uint32_t>> stts;  // (count, delta) runs
   for (size_t i = 0; i < m; i++) {
     size_t si = lo + i;
+    if (d->offsets[si] > d->data.size() ||
+        d->sizes[si] > d->data.size() - d->offsets[si])
+      return cc::set_error(CC_ERR_PARSE, "sample range outside file");
     sizes[i] = d->sizes[si];
     mdat_payload.bytes(d->data.data() + d->offsets[si], d->sizes[si]);
     uint32_t delta;

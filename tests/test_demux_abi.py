@@ -224,3 +224,36 @@ def test_demux_parity_multichunk_co64(built_lib):
             for i in [0, n // 2, n - 1]:
                 pkt, _, _ = d.packet(i)
                 assert pkt == oracle_pkts[i], f"spc={spc} co64={co64} pkt {i}"
+
+
+def test_demux_corrupt_input_never_crashes(built_lib):
+    """Robustness fuzz: truncations and byte-flips of valid containers
+    must produce a clean error or a successful parse — never a crash
+    (a ctypes-level fault would take the process down)."""
+    rng = np.random.default_rng(0xF022)
+    base = mp4_write.write_mp4(
+        [40] * 12, stts=[(12, 512)], ctts=[(1, 512 * o) for o in [2, 4, 1, 1] * 3],
+        timescale=12288, elst_media_time=1024, sync_samples=[1, 5, 9],
+    )
+    cases = 0
+    for _ in range(120):
+        buf = bytearray(base)
+        mode = rng.integers(0, 3)
+        if mode == 0:  # truncate
+            buf = buf[: int(rng.integers(4, len(buf)))]
+        elif mode == 1:  # flip random bytes
+            for _ in range(int(rng.integers(1, 8))):
+                buf[int(rng.integers(0, len(buf)))] = int(rng.integers(0, 256))
+        else:  # corrupt a box size field
+            off = int(rng.integers(0, max(1, len(buf) - 8)))
+            buf[off:off + 4] = int(rng.integers(0, 2**32)).to_bytes(4, "big")
+        data = bytes(buf)
+        try:
+            with hotpath.Demuxer(data) as d:
+                ts = d.timestamps()
+                if len(ts):
+                    d.packet(0)
+        except RuntimeError:
+            pass  # clean error path
+        cases += 1
+    assert cases == 120
