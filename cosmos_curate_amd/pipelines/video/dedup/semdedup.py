@@ -123,6 +123,18 @@ def dedup_cluster(
     return kept, maxv, argi
 
 
+def combine_keep_masks(keep: torch.Tensor, process_group=None) -> torch.Tensor:
+    """AND the per-rank keep masks (each rank pruned only the clusters it
+    owns, leaving the rest True) — the dedup_pipeline.py N>1 combine,
+    RCCL MIN all-reduce on GPU ranks / gloo in the CPU tests."""
+    if process_group is None:
+        return keep
+    ki = keep.int()
+    torch.distributed.all_reduce(ki, op=torch.distributed.ReduceOp.MIN,
+                                 group=process_group)
+    return ki.bool()
+
+
 def semdedup(
     embeddings: torch.Tensor,
     config: SemDedupConfig,

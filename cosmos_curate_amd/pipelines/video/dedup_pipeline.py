@@ -30,6 +30,7 @@ import torch
 
 from cosmos_curate_amd.pipelines.video.dedup.semdedup import (
     SemDedupConfig,
+    combine_keep_masks,
     semdedup,
 )
 
@@ -76,13 +77,7 @@ def dedup(args: argparse.Namespace) -> dict:
                          eps=args.eps)
     t0 = time.perf_counter()
     out = semdedup(e, cfg, process_group=group)
-    keep = out["keep_mask"]
-    if group is not None:
-        # every rank pruned its own cluster subset; AND the masks
-        ki = keep.int()
-        torch.distributed.all_reduce(ki, op=torch.distributed.ReduceOp.MIN,
-                                     group=group)
-        keep = ki.bool()
+    keep = combine_keep_masks(out["keep_mask"], group)
     dt = time.perf_counter() - t0
 
     summary = {

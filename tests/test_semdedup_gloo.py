@@ -62,3 +62,49 @@ def test_kmeans_allreduce_world2():
         e = torch.nn.functional.normalize(torch.from_numpy(shard), dim=1)
         sims = (e @ torch.from_numpy(cent).T).numpy()
         np.testing.assert_array_equal(labels, sims.argmax(axis=1))
+
+
+def _mask_worker(rank, world, port, q):
+    try:
+        os.environ.update(
+            RANK=str(rank), WORLD_SIZE=str(world),
+            MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+        )
+        import torch
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from cosmos_curate_amd.pipelines.video.dedup.semdedup import (
+            combine_keep_masks,
+        )
+
+        # rank r prunes rows r::world (its cluster subset), leaves rest True
+        keep = torch.ones(10, dtype=torch.bool)
+        keep[rank::world] = False
+        combined = combine_keep_masks(keep, dist.group.WORLD)
+        q.put((rank, combined.numpy().tolist()))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, repr(e)))
+
+
+@pytest.mark.timeout(120)
+def test_combine_keep_masks_world2():
+    """dedup_pipeline.py N>1: per-rank keep masks AND together."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_mask_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = {}
+    for _ in ps:
+        rank, val = q.get(timeout=100)
+        results[rank] = val
+    for p in ps:
+        p.join(timeout=30)
+    assert results[0] == results[1] == [False] * 10  # every row pruned by someone
