@@ -191,3 +191,71 @@ def test_mp4_remux_sintel_head_span(golden_dir):
         co, cs = trk.offsets[j], trk.sizes[j]
         assert cs == ss and clip[co:co + cs] == data[so:so + ss]
     assert mp4_demux.annexb_packets(clip, trk)[0][:4] == b"\x00\x00\x00\x01"
+
+
+def test_multicam_driver_input_and_dry_run(tmp_path):
+    """--multi-cam groups UUID session dirs into one task (primary camera
+    first, video_pipe_input.py:238-283); --dry-run prints the stage list."""
+    import argparse
+    import uuid as uuid_mod
+
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import (
+        _setup_parser,
+        build_input_data,
+        split,
+    )
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    root = tmp_path / "sessions"
+    sid = str(uuid_mod.uuid4())
+    for cam in ["rear", "front"]:
+        d = root / sid / cam
+        d.mkdir(parents=True)
+        (d / "v.nv12").write_bytes(
+            raw_backend.make_synthetic_clip(60, 32, 48, 30, seed=1))
+    (root / "not-a-session").mkdir()  # ignored: not a UUID name
+
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    args = p.parse_args([
+        "--input-video-path", str(root), "--output-clip-path",
+        str(tmp_path / "out"), "--multi-cam",
+        "--primary-camera-keyword", "front", "--dry-run",
+    ])
+    tasks = build_input_data(args)
+    assert len(tasks) == 1 and tasks[0].session_id == sid
+    assert len(tasks[0].videos) == 2
+    assert "front" in str(tasks[0].videos[0].input_video)  # primary first
+
+    summary = split(args)
+    assert summary["dry_run"] and summary["num_input_videos"] == 1
+
+    # transnetv2 + multicam is rejected loudly (reference :213)
+    args2 = p.parse_args([
+        "--input-video-path", str(root), "--output-clip-path",
+        str(tmp_path / "o2"), "--multi-cam",
+        "--splitting-algorithm", "transnetv2",
+    ])
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="fixed-stride"):
+        build_input_data(args2)
+
+
+def test_driver_flag_aliases():
+    """Reference flag spellings parse to the same dests."""
+    import argparse
+
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import _setup_parser
+
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    a = p.parse_args([
+        "--input-video-path", "/i", "--output-clip-path", "/o",
+        "--fixed-stride-min-clip-length-s", "5",
+        "--no-generate-embeddings",
+        "--motion-filter", "score-only",
+    ])
+    assert a.fixed_stride_min_clip_length == 5.0
+    assert a.generate_embeddings is False
+    assert a.motion_filter == "score-only"
