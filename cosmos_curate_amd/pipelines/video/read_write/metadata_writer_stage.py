@@ -89,16 +89,22 @@ class ClipWriterStage(CuratorStage):
             self._timer.reinit(self, task.get_major_size())
             with self._timer.time_process():
                 emb_rows: list[tuple[str, np.ndarray]] = []
-                for video in task.videos:
+                multicam = len(task.videos) > 1
+                for cam_idx, video in enumerate(task.videos):
                     for clip in video.clips:
-                        cid = str(clip.uuid)
+                        # multicam cameras share clip uuids (aligned spans);
+                        # per-camera outputs get a camera prefix so files
+                        # don't collide (AV writer camera_id convention)
+                        cid = (f"cam{cam_idx}_{clip.uuid}" if multicam
+                               else str(clip.uuid))
                         if self._upload_clips and clip.encoded_data:
                             payload = clip.encoded_data.resolve()
                             (self._output_path / "clips" / f"{cid}.bin").write_bytes(
                                 bytes(payload)
                             )
                         meta = {
-                            "uuid": cid,
+                            "uuid": str(clip.uuid),
+                            "camera_index": cam_idx if multicam else None,
                             "source_video": clip.source_video,
                             "span": list(clip.span),
                             "duration": clip.duration,

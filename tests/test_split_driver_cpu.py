@@ -259,3 +259,48 @@ def test_driver_flag_aliases():
     assert a.fixed_stride_min_clip_length == 5.0
     assert a.generate_embeddings is False
     assert a.motion_filter == "score-only"
+
+
+def test_multicam_split_end_to_end_cpu(tmp_path):
+    """Full multicam split on CPU (no embeddings): both cameras get the
+    same clip spans/uuids and both cameras' clips are written."""
+    import argparse
+    import json
+    import uuid as uuid_mod
+
+    from cosmos_curate_amd.core.interfaces import SequentialRunner
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import (
+        _setup_parser,
+        split,
+    )
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    root = tmp_path / "sessions"
+    sid = str(uuid_mod.uuid4())
+    for cam in ["front", "rear"]:
+        d = root / sid / cam
+        d.mkdir(parents=True)
+        (d / "v.nv12").write_bytes(
+            raw_backend.make_synthetic_clip(600, 32, 48, 30, seed=3))
+    out = tmp_path / "out"
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    args = p.parse_args([
+        "--input-video-path", str(root), "--output-clip-path", str(out),
+        "--multi-cam", "--no-embeddings",
+    ])
+    summary = split(args, runner=SequentialRunner())
+    assert summary["num_input_videos"] == 2  # 2 cameras in 1 session
+    assert summary["num_clips"] == 4  # 20 s x 2 cams -> 2 clips each
+    assert summary["num_clips_with_errors"] == 0
+    metas = sorted((out / "metas" / "v0").glob("*.json"))
+    assert len(metas) == 4  # per-camera files (cam prefix), no collisions
+    spans = {}
+    for m in metas:
+        meta = json.loads(m.read_text())
+        spans.setdefault(meta["span"][0], []).append(meta["uuid"])
+    # multicam cameras share the SAME clip uuid per span
+    # (clip_extraction_stages.py:654-661)
+    assert len(spans) == 2
+    for uuids in spans.values():
+        assert len(uuids) == 2 and uuids[0] == uuids[1]
