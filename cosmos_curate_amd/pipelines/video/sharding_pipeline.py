@@ -108,6 +108,7 @@ def write_webdataset_shards(
     target_size_bytes: int = 256 * 1024 * 1024,
     *,
     drop_small_shards: bool = False,
+    min_clips_per_tar: int = MIN_CLIPS_PER_TAR_DEFAULT,
 ) -> dict:
     """Bin + pack + write tars; returns the shard index (also on disk)."""
     out_root = pathlib.Path(output_path)
@@ -118,7 +119,7 @@ def write_webdataset_shards(
         bin_dir.mkdir(parents=True, exist_ok=True)
         shards = []
         for i, pack in enumerate(
-            group_samples_by_size(group, target_size_bytes, drop_small_shards=drop_small_shards)
+            group_samples_by_size(group, target_size_bytes, drop_small_shards=drop_small_shards, min_clips_per_tar=min_clips_per_tar)
         ):
             tar_path = bin_dir / f"shard_{i:05d}.tar"
             with tarfile.open(tar_path, "w") as tar:
@@ -169,13 +170,42 @@ def load_samples_from_split_output(split_output: str) -> list[ClipSample]:
     return samples
 
 
+def filter_samples_by_semantic_dedup(
+    samples: list[ClipSample], dedup_path: str,
+) -> list[ClipSample]:
+    """Drop samples the dedup pipeline pruned
+    (video_pipe_input.py:514-560 semantics over dedup_pipeline.py's
+    dedup_results.parquet: keep == False rows are removed)."""
+    import pathlib
+
+    import pyarrow.parquet as pq
+
+    files = sorted(pathlib.Path(dedup_path).rglob("dedup_results.parquet"))
+    if not files:
+        msg = f"no dedup_results.parquet under {dedup_path}"
+        raise FileNotFoundError(msg)
+    remove: set[str] = set()
+    for f in files:
+        t = pq.read_table(f, columns=["id", "keep"])
+        for cid, keep in zip(t.column("id").to_pylist(),
+                             t.column("keep").to_pylist()):
+            if not keep:
+                remove.add(str(cid))
+    return [s for s in samples if str(s.uuid) not in remove]
+
+
 def shard(args: argparse.Namespace) -> dict:
     """sharding_pipeline.py:194 driver shape."""
     samples = load_samples_from_split_output(args.input_clip_path)
+    if getattr(args, "input_semantic_dedup_path", None):
+        samples = filter_samples_by_semantic_dedup(
+            samples, args.input_semantic_dedup_path)
     return write_webdataset_shards(
         samples, args.output_shard_path,
         target_size_bytes=args.target_shard_size_mb * 1024 * 1024,
         drop_small_shards=args.drop_small_shards,
+        min_clips_per_tar=getattr(args, "min_clips_per_tar",
+                                  MIN_CLIPS_PER_TAR_DEFAULT),
     )
 
 
@@ -185,6 +215,11 @@ def cli_run_shard(argv: list[str] | None = None) -> dict:
     p.add_argument("--output-shard-path", required=True)
     p.add_argument("--target-shard-size-mb", type=int, default=256)
     p.add_argument("--drop-small-shards", action="store_true")
+    p.add_argument("--min-clips-per-tar", type=int, default=MIN_CLIPS_PER_TAR_DEFAULT)
+    p.add_argument("--input-semantic-dedup-path", default=None,
+                   help="dedup pipeline output dir; pruned clips are "
+                   "excluded from the shards (reference "
+                   "--input-semantic-dedup-path)")
     args = p.parse_args(argv)
     index = shard(args)
     total = sum(s["clips"] for shards in index.values() for s in shards)

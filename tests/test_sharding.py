@@ -80,3 +80,40 @@ def test_shard_from_split_output(tmp_path):
     ])
     total = sum(s["clips"] for shards in index.values() for s in shards)
     assert total == 2  # 20 s video -> two 10 s clips
+
+
+def test_shard_consumes_dedup_results(tmp_path):
+    """split -> dedup results -> shard: pruned clips excluded
+    (video_pipe_input.py:514-560 semantics)."""
+    import argparse
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from cosmos_curate_amd.core.interfaces import SequentialRunner
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import _setup_parser, split
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    inp = tmp_path / "in"
+    inp.mkdir()
+    (inp / "v.nv12").write_bytes(raw_backend.make_synthetic_clip(600, 64, 96, 30, seed=1))
+    out = tmp_path / "split"
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    split(p.parse_args(["--input-video-path", str(inp), "--output-clip-path",
+                        str(out), "--no-embeddings"]), runner=SequentialRunner())
+    metas = sorted((out / "metas" / "v0").glob("*.json"))
+    ids = [json.loads(m.read_text())["uuid"] for m in metas]
+    assert len(ids) == 2
+    dd = tmp_path / "dedup"
+    dd.mkdir()
+    pq.write_table(pa.table({"id": ids, "cluster": [0, 0],
+                             "keep": [True, False]}),
+                   dd / "dedup_results.parquet")
+    index = shp.cli_run_shard([
+        "--input-clip-path", str(out), "--output-shard-path", str(tmp_path / "sh"),
+        "--target-shard-size-mb", "1",
+        "--input-semantic-dedup-path", str(dd),
+    ])
+    total = sum(s["clips"] for shards in index.values() for s in shards)
+    assert total == 1  # the pruned clip is excluded
