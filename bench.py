@@ -4,9 +4,12 @@ Measures BASELINE.json's metric ("clips/sec + embedded frames/sec, 1080p30
 H.264") on the workload of BASELINE configs[1] — the largest single-GPU
 configuration: per clip, 21 sampled 1080p NV12 frames (10 s @ 30 fps
 sampled at 2 fps with the endpoint rule) -> fused NV12->RGB + bilinear
-resize to 224 (HIP) -> fused CLIP preprocess to bf16 NCHW (HIP) ->
-CLIP-ViT-B/32 forward with custom MFMA bf16 GEMMs -> mean-pooled
-L2-normalized clip embedding (copied to host).
+resize (HIP) -> fused normalize + patch-extraction straight into the
+GEMM A layout (HIP) -> ViT forward entirely on hand-written kernels
+(MFMA bf16 GEMMs with fused epilogues, fused LayerNorm, LDS-resident /
+streaming attention by sequence length) -> mean-pooled L2-normalized
+clip embedding (copied to host).  --model selects the tower: vit_b32
+(flagship), vit_l14, siglip_l16_256.
 
 The timed region starts with NV12 surfaces already resident in HBM (tier
 contract; H.264 decode itself needs librocdecode, absent from this image —
