@@ -304,3 +304,37 @@ def test_multicam_split_end_to_end_cpu(tmp_path):
     assert len(spans) == 2
     for uuids in spans.values():
         assert len(uuids) == 2 and uuids[0] == uuids[1]
+
+
+def test_run_pipeline_unified_entry(tmp_path):
+    """run_pipeline CLI-mode + config-mode dispatch (reference
+    run_pipeline.py:17-27 semantics)."""
+    import json as json_mod
+
+    from cosmos_curate_amd.pipelines.video.run_pipeline import main
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    inp = tmp_path / "in"
+    inp.mkdir()
+    (inp / "v.nv12").write_bytes(raw_backend.make_synthetic_clip(300, 32, 48, 30, seed=2))
+
+    # CLI mode
+    s1 = main(["split", "--input-video-path", str(inp),
+               "--output-clip-path", str(tmp_path / "o1"), "--no-embeddings"])
+    assert s1["num_clips"] == 1
+
+    # config mode (json)
+    cfg = tmp_path / "job.json"
+    cfg.write_text(json_mod.dumps({
+        "pipeline": "split",
+        "args": {"input_video_path": str(inp),
+                 "output_clip_path": str(tmp_path / "o2"),
+                 "no_embeddings": True},
+    }))
+    s2 = main([str(cfg)])
+    assert s2["num_clips"] == 1
+
+    import pytest as _pytest
+
+    with _pytest.raises(SystemExit):
+        main(["nope"])
