@@ -56,13 +56,14 @@ class SequentialRunner(RunnerInterface):
         _model_weights_prefix: str = "",
         _execution_mode: str = "AUTO",
     ) -> list[T] | None:
-        stages: list[CuratorStage] = [spec.stage for spec in stage_specs]
-        for stage in stages:
-            stage.stage_setup_on_node()
-            stage.stage_setup()
         tasks: list[PipelineTask] = list(input_tasks)
         for spec in stage_specs:
             stage = spec.stage
+            # setup just before the stage runs (actor-lifetime shape:
+            # xenna builds each stage's actors when the stage starts) —
+            # keeps heavyweight model setup from front-loading memory
+            stage.stage_setup_on_node()
+            stage.stage_setup()
             attempts = max(1, spec.num_run_attempts_python)
             result: list[PipelineTask] | None = None
             for attempt in range(attempts):
