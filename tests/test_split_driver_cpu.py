@@ -334,6 +334,18 @@ def test_run_pipeline_unified_entry(tmp_path):
     s2 = main([str(cfg)])
     assert s2["num_clips"] == 1
 
+    # config mode (yaml)
+    ycfg = tmp_path / "job.yaml"
+    ycfg.write_text(
+        "pipeline: split\n"
+        "args:\n"
+        f"  input_video_path: {inp}\n"
+        f"  output_clip_path: {tmp_path / 'o3'}\n"
+        "  no_embeddings: true\n"
+    )
+    s3 = main([str(ycfg)])
+    assert s3["num_clips"] == 1
+
     import pytest as _pytest
 
     with _pytest.raises(SystemExit):
