@@ -81,3 +81,29 @@ def test_signature_format_matches_reference():
 def test_bad_sample_rate_raises():
     with pytest.raises(ValueError):
         sample_closest(np.arange(4, dtype=np.float32), -1.0)
+
+
+def test_sample_closest_extreme_ratios():
+    """find_closest/sample_closest at extreme fps ratios: product ==
+    oracle for upsampling (target >> source), near-equal rates, and
+    very sparse sampling (decoder_utils.py:356-453 semantics)."""
+    import numpy as np
+
+    from cosmos_curate_amd.pipelines.video.utils import decoder_utils as du
+    from oracle import sampling as osamp
+
+    rng = np.random.default_rng(0xA7E5)
+    cases = []
+    for _ in range(40):
+        fps = float(rng.choice([10, 23.976, 24, 29.97, 30, 59.94, 120]))
+        n = int(rng.integers(2, 400))
+        target = float(rng.choice([0.1, 0.5, 1, 2, 5, fps, fps * 2, 240]))
+        cases.append((fps, n, target))
+    cases += [(30.0, 1, 2.0), (30.0, 2, 240.0), (120.0, 300, 0.1)]
+    for fps, n, target in cases:
+        ts = (np.arange(n) / fps).astype(np.float32)
+        gi, gc, gs = du.sample_closest(ts, sample_rate=target)
+        oi, oc, os_ = osamp.sample_closest(ts, target)
+        np.testing.assert_array_equal(gi, oi, err_msg=f"{fps=} {n=} {target=}")
+        np.testing.assert_array_equal(gc, oc)
+        assert gs == os_
