@@ -106,4 +106,4 @@ def test_sample_closest_extreme_ratios():
         oi, oc, os_ = osamp.sample_closest(ts, target)
         np.testing.assert_array_equal(gi, oi, err_msg=f"{fps=} {n=} {target=}")
         np.testing.assert_array_equal(gc, oc)
-        assert gs == os_
+        np.testing.assert_array_equal(gs, os_)
