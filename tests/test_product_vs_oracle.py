@@ -107,3 +107,27 @@ def test_sample_closest_extreme_ratios():
         np.testing.assert_array_equal(gi, oi, err_msg=f"{fps=} {n=} {target=}")
         np.testing.assert_array_equal(gc, oc)
         np.testing.assert_array_equal(gs, os_)
+
+
+def test_fixed_stride_spans_randomized_vs_oracle():
+    """Randomized stride/len/min combos: product span math == oracle
+    (clip_extraction_stages.py:512-551), including overlap (stride <
+    len), sparse (stride > len), and fractional-second configurations."""
+    import numpy as np
+
+    from cosmos_curate_amd.pipelines.video.clipping.clip_extraction_stages import (
+        _make_clip_uuids,
+        _make_spans_fixed_stride,
+    )
+    from oracle import spans as ospans
+
+    rng = np.random.default_rng(0x5A9)
+    for _ in range(60):
+        end = float(rng.uniform(0.5, 120.0))
+        clip_len = float(rng.uniform(0.5, 30.0))
+        stride = float(rng.uniform(0.25, 30.0))
+        min_len = float(rng.uniform(0.0, clip_len))
+        got = _make_spans_fixed_stride(0.0, end, clip_len, stride, min_len)
+        want = ospans.make_spans_fixed_stride(0.0, end, clip_len, stride, min_len)
+        assert got == want, (end, clip_len, stride, min_len)
+        assert _make_clip_uuids("s", got) == ospans.make_clip_uuids("s", want)
