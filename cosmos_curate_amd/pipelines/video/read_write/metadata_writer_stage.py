@@ -18,6 +18,7 @@ clips/, metas/v0/, <embedding>_embd/ parquet).
 from __future__ import annotations
 
 import json
+import os
 import pathlib
 import time
 
@@ -52,6 +53,10 @@ class ClipWriterStage(CuratorStage):
         self._verbose = verbose
         self._log_stats = log_stats
         self._chunk_counter = 0
+        # chunk files are rank-namespaced: in a WorkerPoolRunner run one
+        # process per GPU shares the output dir, and a bare per-instance
+        # counter would overwrite other ranks' chunk_000000.parquet
+        self._rank = int(os.environ.get("RANK", os.environ.get("LOCAL_RANK", "0")))
 
     @property
     def resources(self) -> CuratorStageResource:
@@ -79,7 +84,7 @@ class ClipWriterStage(CuratorStage):
         dest = (
             self._output_path
             / f"{self._embedding_algorithm}_embd"
-            / f"chunk_{self._chunk_counter:06d}.parquet"
+            / f"chunk_{self._rank}_{self._chunk_counter:06d}.parquet"
         )
         pq.write_table(table, dest)
         self._chunk_counter += 1
@@ -103,6 +108,11 @@ class ClipWriterStage(CuratorStage):
                                 bytes(payload)
                             )
                         meta = {
+                            # "id" is the file/sample id (cam-prefixed for
+                            # multicam); downstream consumers (sharding,
+                            # dedup matching) key on it, not on the bare
+                            # uuid, which multicam cameras share
+                            "id": cid,
                             "uuid": str(clip.uuid),
                             "camera_index": cam_idx if multicam else None,
                             "source_video": clip.source_video,

@@ -147,7 +147,12 @@ def load_samples_from_split_output(split_output: str) -> list[ClipSample]:
     samples = []
     for meta_path in sorted((root / "metas" / "v0").glob("*.json")):
         meta = json.loads(meta_path.read_text())
-        payload_path = root / "clips" / f"{meta['uuid']}.bin"
+        # sample id = meta filename stem = ClipWriterStage's file id
+        # (cam-prefixed for multicam, bare uuid otherwise); matches the
+        # clips/<id>.bin payload name and the embedding-parquet "id"
+        # column that dedup_results.parquet keys on
+        sample_id = meta.get("id", meta_path.stem)
+        payload_path = root / "clips" / f"{sample_id}.bin"
         if not payload_path.exists():
             continue
         payload = payload_path.read_bytes()
@@ -164,7 +169,7 @@ def load_samples_from_split_output(split_output: str) -> list[ClipSample]:
             md = extract_video_metadata(payload)
             n, h, w, fps = md.num_frames, md.height, md.width, md.fps
         samples.append(
-            ClipSample(uuid=meta["uuid"], payload=payload, width=w, height=h,
+            ClipSample(uuid=sample_id, payload=payload, width=w, height=h,
                        framerate=fps, num_frames=n, metadata=meta)
         )
     return samples

@@ -117,3 +117,55 @@ def test_shard_consumes_dedup_results(tmp_path):
     ])
     total = sum(s["clips"] for shards in index.values() for s in shards)
     assert total == 1  # the pruned clip is excluded
+
+
+def test_multicam_split_to_shard(tmp_path):
+    """Multicam split -> shard: cam-prefixed sample ids resolve payloads
+    and match dedup ids (previously every multicam clip was
+    silently skipped because the loader looked up clips/<bare-uuid>.bin)."""
+    import argparse
+    import uuid as uuid_mod
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from cosmos_curate_amd.core.interfaces import SequentialRunner
+    from cosmos_curate_amd.pipelines.video.splitting_pipeline import _setup_parser, split
+    from cosmos_curate_amd.pipelines.video.utils import raw_backend
+
+    root = tmp_path / "sessions"
+    sid = str(uuid_mod.uuid4())
+    for cam in ["front", "rear"]:
+        d = root / sid / cam
+        d.mkdir(parents=True)
+        (d / "v.nv12").write_bytes(
+            raw_backend.make_synthetic_clip(600, 32, 48, 30, seed=3))
+    out = tmp_path / "split"
+    p = argparse.ArgumentParser()
+    _setup_parser(p)
+    split(p.parse_args([
+        "--input-video-path", str(root), "--output-clip-path", str(out),
+        "--multi-cam", "--no-embeddings",
+    ]), runner=SequentialRunner())
+
+    samples = shp.load_samples_from_split_output(str(out))
+    # 2 cams x 2 clips, all loaded (cam-prefixed payload names resolve)
+    assert len(samples) == 4
+    assert all(s.uuid.startswith("cam") for s in samples)
+    assert {s.metadata["camera_index"] for s in samples} == {0, 1}
+
+    # dedup results keyed by the same cam-prefixed ids: prune one camera's
+    # clip and verify exactly that sample drops out
+    ids = sorted(s.uuid for s in samples)
+    dd = tmp_path / "dedup"
+    dd.mkdir()
+    pq.write_table(pa.table({"id": ids, "cluster": [0] * 4,
+                             "keep": [False, True, True, True]}),
+                   dd / "dedup_results.parquet")
+    kept = shp.filter_samples_by_semantic_dedup(samples, str(dd))
+    assert sorted(s.uuid for s in kept) == ids[1:]
+
+    index = shp.write_webdataset_shards(kept, str(tmp_path / "shards"),
+                                        target_size_bytes=1 << 20)
+    total = sum(s["clips"] for shards in index.values() for s in shards)
+    assert total == 3
