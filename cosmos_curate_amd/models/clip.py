@@ -102,6 +102,31 @@ class _CLIPImageEmbeddings(torch.nn.Module):
         )
         return out
 
+    def _resize_center_crop_u8(self, frames_dev_u8: torch.Tensor) -> torch.Tensor:
+        """(N,H,W,3) u8 device -> (N,image,image,3) u8: bicubic resize of
+        the shorter side to cfg.image + center crop (torchvision
+        Resize(BICUBIC)+CenterCrop chain, reference clip.py:48-56), on
+        device via cc_resize_bicubic_u8."""
+        import ctypes  # noqa: F401 — hotpath ctypes setup already done
+
+        lib = hotpath.require_gpu()
+        n, h, w, _ = frames_dev_u8.shape
+        size = self.tower.cfg.image
+        scale = size / min(h, w)
+        rh, rw = max(size, round(h * scale)), max(size, round(w * scale))
+        out = torch.empty((n, rh, rw, 3), dtype=torch.uint8,
+                          device=frames_dev_u8.device)
+        stream = torch.cuda.current_stream(frames_dev_u8.device).cuda_stream
+        hotpath.check(
+            lib.cc_resize_bicubic_u8(
+                frames_dev_u8.contiguous().data_ptr(), n, h, w,
+                out.data_ptr(), rh, rw, stream,
+            )
+        )
+        top = (rh - size) // 2
+        left = (rw - size) // 2
+        return out[:, top:top + size, left:left + size, :].contiguous()
+
     @torch.no_grad()
     def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
         if isinstance(images, np.ndarray):
@@ -110,11 +135,15 @@ class _CLIPImageEmbeddings(torch.nn.Module):
         if (
             images.dtype == torch.uint8 and images.ndim == 4
             and images.shape[-1] == 3
-            and images.shape[1] == self.tower.cfg.image
-            and images.shape[2] == self.tower.cfg.image
         ):
+            images = images.to(self.device)
+            if (images.shape[1] != self.tower.cfg.image
+                    or images.shape[2] != self.tower.cfg.image):
+                # target_res=-1 runs hand source-resolution frames here;
+                # resize+crop on device first (docstring contract above)
+                images = self._resize_center_crop_u8(images)
             n = images.shape[0]
-            patches = self.preprocess_patches_u8(images.to(self.device))
+            patches = self.preprocess_patches_u8(images)
             return self.tower(patches=patches, n=n)
         pixels = images.to(self.device, dtype=torch.bfloat16)
         return self.tower(pixels)

@@ -53,6 +53,7 @@ class ClipFrameCreationStage(CuratorStage):
         *,
         max_frames: int | None = None,
         min_frames: int | None = None,
+        target_res: tuple[int, int] | None = None,
         verbose: bool = False,
         log_stats: bool = False,
     ) -> None:
@@ -60,6 +61,9 @@ class ClipFrameCreationStage(CuratorStage):
         self._target_fps = target_fps
         self._max_frames = max_frames
         self._min_frames = min_frames
+        # re-extraction must regenerate frames at the SAME resolution the
+        # extraction stage produced, or the embedder sees mixed sizes
+        self._target_res = target_res if target_res is not None else (-1, -1)
         self._verbose = verbose
         self._log_stats = log_stats
 
@@ -92,6 +96,7 @@ class ClipFrameCreationStage(CuratorStage):
                 raw = bytes(data) if not isinstance(data, bytes) else data
                 frames = extract_frames(
                     raw, sample_rate_fps=regen_fps,
+                    target_res=self._target_res,
                     to_host=isinstance(frames, np.ndarray),
                 )
         if self._max_frames is not None and len(frames) > self._max_frames:

@@ -195,7 +195,12 @@ def _assemble_stages(args: argparse.Namespace) -> list[CuratorStage | CuratorSta
             log_stats=True,
         ))
     if args.generate_embeddings:
-        stages.append(ClipFrameCreationStage(target_fps=args.target_clip_fps, log_stats=True))
+        res = args.clip_extraction_target_res
+        stages.append(ClipFrameCreationStage(
+            target_fps=args.target_clip_fps,
+            target_res=(res, res) if res > 0 else (-1, -1),
+            log_stats=True,
+        ))
         stages.append(ClipEmbeddingStage(log_stats=True))
     stages.append(ClipWriterStage(args.output_clip_path, log_stats=True))
     return stages

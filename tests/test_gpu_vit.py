@@ -438,3 +438,35 @@ def test_gemm_tanh_gelu_epilogue(lib, M, N, K):
     y = a.float().cpu() @ b.float().cpu().T + bias.cpu()
     y = torch.nn.functional.gelu(y, approximate="tanh") + res.float().cpu()
     torch.testing.assert_close(out.float().cpu(), y, rtol=2e-2, atol=8e-2)
+
+
+def test_clip_non224_input_resized_on_device(lib):
+    """target_res=-1 runs hand source-resolution frames to the embedder;
+    __call__ must resize+center-crop on device (previously the
+    u8 non-224 path hit a reshape failure and every clip errored).
+    Parity: torchvision-equivalent CPU chain (bicubic shorter-side resize
+    + center crop) -> fp32 oracle, cosine >= 0.999."""
+    from cosmos_curate_amd.models.clip import CLIPImageEmbeddings
+    from cosmos_curate_amd.models.clip_weights import make_clip_vit_b32_weights
+    from oracle import vit as oracle_vit
+    from oracle.color import clip_preprocess, resize_bicubic_u8
+
+    model = CLIPImageEmbeddings()
+    model.setup()
+    rng = np.random.default_rng(7)
+    frames = rng.integers(0, 256, size=(2, 240, 320, 3), dtype=np.uint8)
+    emb = model(frames)
+    assert emb.shape == (2, 512)
+    norms = torch.linalg.vector_norm(emb, dim=-1).cpu()
+    torch.testing.assert_close(norms, torch.ones(2), rtol=1e-3, atol=1e-3)
+
+    # CPU reference of the same chain: shorter side -> 224 bicubic, center crop
+    scale = 224 / 240
+    rh, rw = max(224, round(240 * scale)), max(224, round(320 * scale))
+    resized = resize_bicubic_u8(frames, rh, rw)
+    top, left = (rh - 224) // 2, (rw - 224) // 2
+    cropped = resized[:, top:top + 224, left:left + 224, :]
+    ref = oracle_vit.build_reference_clip_vision(make_clip_vit_b32_weights())
+    want = oracle_vit.embed_frames_fp32(ref, clip_preprocess(cropped))
+    cos = np.sum(want * emb.cpu().numpy(), axis=1)
+    assert np.all(cos >= 0.999), cos

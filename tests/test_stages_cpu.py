@@ -250,3 +250,41 @@ def test_timestamps_integration_real_mp4():
         assert "timestamps" not in v2.errors and v2.timestamps is not None
         out = run_pipeline([t], [FixedStrideExtractorStage()], runner=SequentialRunner())
         assert len(out[0].video.clips) == 3
+
+
+def test_frame_creation_reextraction_passes_target_res(monkeypatch):
+    """min_frames re-extraction must regenerate at the stage's target_res
+    (it previously omitted it, regenerating source-resolution
+    frames that failed downstream)."""
+    import numpy as np
+
+    from cosmos_curate_amd.pipelines.video.embedding import clip_stages
+    from cosmos_curate_amd.pipelines.video.utils.data_model import (
+        Clip,
+        LazyData,
+        SplitPipeTask,
+        Video,
+    )
+    from cosmos_curate_amd.pipelines.video.utils.decoder_utils import (
+        FrameExtractionPolicy,
+        FrameExtractionSignature,
+    )
+
+    seen_kwargs = {}
+
+    def fake_extract_frames(data, *, sample_rate_fps, target_res=(-1, -1), to_host=False):
+        seen_kwargs["target_res"] = target_res
+        return np.zeros((8, target_res[0], target_res[1], 3), dtype=np.uint8)
+
+    monkeypatch.setattr(clip_stages, "extract_frames", fake_extract_frames)
+
+    sig = FrameExtractionSignature(FrameExtractionPolicy.sequence, 2.0).to_str()
+    few = np.zeros((2, 64, 64, 3), dtype=np.uint8)
+    clip = Clip(uuid="u", source_video="s", span=(0.0, 10.0))
+    clip.extracted_frames = LazyData(value={sig: few}, nbytes=few.nbytes)
+    clip.encoded_data = LazyData(value=b"payload", nbytes=7)
+    stage = clip_stages.ClipFrameCreationStage(
+        target_fps=2.0, min_frames=8, target_res=(64, 64))
+    stage.process_data([SplitPipeTask(videos=[Video(input_video="v", clips=[clip])])])
+    assert seen_kwargs.get("target_res") == (64, 64)
+    assert len(clip.clip_embedding_frames.resolve()) == 8
