@@ -1,0 +1,767 @@
+// V13/V14: register-staged 256x256 GEMM pipeline (round-2 experiment).
+//
+// Why: the production t256 body (csrc/cc_gemm.hip k_gemm_bf16_t256)
+// carries one full `s_waitcnt vmcnt(0)` drain per K-tile because counted
+// vmcnt waits on global_load_lds were measured NONDETERMINISTIC on this
+// pool (profiles/r01_t256_det2.log — LDS-DMA completions retire out of
+// order under load).  That drain exposes raw HBM latency once per K-tile
+// with 1 WG/CU and nothing to cover it; at the short-K ViT shapes
+// (KT=12) it plus prologue/epilogue is most of the gap to the >=40%
+// MFMA target (BENCH_r01 frac 0.29).
+//
+// Fix tried here: stage through REGISTERS — plain global loads into
+// VGPRs, ds_write into LDS.  Plain loads are scoreboarded per register
+// by hipcc (in-order vmcnt semantics hold for VMEM-to-VGPR returns), so
+// the compiler inserts precise counted waits and the K-loop carries NO
+// global drain at all.  Schedule per K-tile t (buf = t&1, 4 phases, one
+// leading barrier per phase — the v11 discipline, hazards separated by
+// >=1 barrier via the pre-barrier-issue property):
+//   ph1: ds_read B(t) all + A q0; ds_write A(t+1) (4 dsw -> buf^1)
+//   ph2: ds_read A q1;            ds_write B(t+1) (4 dsw);
+//        global loads A(t+2) -> aregs (4)
+//   ph3: ds_read A q2;            global loads B(t+2) -> bregs (4)
+//   ph4: ds_read A q3
+// Hazard check (all 1+ barrier separated by pre-barrier issue):
+//   WAR buf^1: tile t-1's last reads of buf^1 issue at its ph4
+//     pre-barrier; writes issue at tile t ph1 pre-barrier (1 barrier).
+//   RAW next tile: writes done by ph2 pre-barrier; next-tile reads
+//     issue at ph4 pre-barrier / t+1 ph1 (2 barriers).
+//   Register WAR: aregs consumed by dsw at ph1, reloaded ph2; bregs
+//     consumed ph2, reloaded ph3 (compiler-enforced anyway).
+// V14 = V13 + persistent multi-tile outer loop: one WG computes several
+// output tiles; the next tile's global loads issue BEFORE the epilogue
+// stores, so the prologue latency hides under the C writeback.
+//
+// Verification: sync structure is NEW -> two-lane discipline: refcheck
+// vs production at small sizes + multi-run determinism screen at the
+// big ViT grids (tools/gemm_v13_screen.py).
+
+#include <hip/hip_runtime.h>
+
+#include "../cosmos_curate_amd/csrc/cc_common.hpp"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int BM = 256, BN = 256, BK = 64;
+constexpr int WM = 128, WN = 64;  // per-wave output (2M x 4N wave grid)
+constexpr int FRAG = 16;
+constexpr int MFR = WM / FRAG;  // 8
+constexpr int NFR = WN / FRAG;  // 4
+
+__device__ __forceinline__ unsigned short bf16_rne(float v) {
+  union { float f; unsigned int u; } cv{v};
+  return (unsigned short)((cv.u + 0x7fffu + ((cv.u >> 16) & 1)) >> 16);
+}
+
+__device__ __forceinline__ bf16x8 frag_read(const __bf16* tile, int row,
+                                            int k16) {
+  int slot = k16 ^ (row & 7);
+  return *(const bf16x8*)(tile + (long)row * BK + slot * 8);
+}
+
+// ---- register staging ----
+// A wave owns 32 rows of each 256-row operand tile per K-tile, as 4
+// groups of 8 rows: groups 0,1 = half0 rows wid*16..+16, groups 2,3 =
+// half1 rows 128+wid*16..+16.  Lane covers (row = grp + (lane>>3),
+// slot = lane&7): one bf16x8 (16 B) per group -> coalesced 128 B per
+// 8 lanes, identical addressing to the glds path's source.
+__device__ __forceinline__ long stage_row(int wid, int lane, int g) {
+  return (long)((g >> 1) * 128 + wid * 16 + (g & 1) * 8 + (lane >> 3));
+}
+
+#define LOAD_REGS(dst, src, ld, base, limit, k0)                            \
+  _Pragma("unroll") for (int g = 0; g < 4; g++) {                           \
+    long grow = (base) + stage_row(wid, lane, g);                           \
+    grow = grow < 0 ? 0 : (grow >= (limit) ? (limit)-1 : grow);             \
+    dst[g] = *(const bf16x8*)((src) + grow * (ld) + (k0) + slot * 8);       \
+  }
+
+#define WRITE_REGS(srcreg, ldsbase)                                         \
+  _Pragma("unroll") for (int g = 0; g < 4; g++) {                           \
+    long lrow = stage_row(wid, lane, g);                                    \
+    *(bf16x8*)((ldsbase) + lrow * BK + ((long)(slot ^ (lrow & 7))) * 8) =   \
+        srcreg[g];                                                          \
+  }
+
+template <int PERSIST>  // 0: one tile per WG; 1: grid-strided tiles
+__global__ __launch_bounds__(512, 1) void k_gemm_v13(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // 128 KiB
+#define A13(b) (lds + (b) * (BM * BK))
+#define B13(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const int slot = lane & 7;
+  const long KT = K / BK;  // caller guarantees KT >= 2
+
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+
+  bf16x8 aregs[4], bregs[4];
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2];
+
+  const int tiles = PERSIST ? (nwg + (int)gridDim.x - 1) / (int)gridDim.x : 1;
+  for (int rep = 0; rep < tiles; rep++) {
+    int orig = PERSIST ? (int)blockIdx.x + rep * (int)gridDim.x
+                       : (int)blockIdx.x;
+    if (orig >= nwg) break;
+    if (do_remap) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    const long bm = (long)(orig / nbx) * BM;
+    const long bn = (long)(orig % nbx) * BN;
+
+    f32x4 acc[MFR][NFR] = {};
+
+    // prologue: tile 0 via regs (on rep>0 the loads were issued before
+    // the previous epilogue, hiding their latency under the C stores)
+    if (rep == 0) {
+      LOAD_REGS(aregs, A, K, bm, M, 0);
+      LOAD_REGS(bregs, B, K, bn, N, 0);
+    }
+    WRITE_REGS(aregs, A13(0));
+    WRITE_REGS(bregs, B13(0));
+    if (KT > 1) {
+      LOAD_REGS(aregs, A, K, bm, M, BK);
+      LOAD_REGS(bregs, B, K, bn, N, BK);
+    }
+    __syncthreads();  // publish buf0 (drains the ds_writes)
+
+#define PHASE_MFMA13(q)                                                     \
+  do {                                                                      \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define READ_A13(At, q)                                                     \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =             \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B13(Bt)                                                        \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+    for (long t = 0; t < KT; ++t) {
+      const __bf16* At = A13(t & 1);
+      const __bf16* Bt = B13(t & 1);
+      __bf16* Aw = A13((t & 1) ^ 1);
+      __bf16* Bw = B13((t & 1) ^ 1);
+      const bool stage_next = t + 1 < KT;    // write t+1 into buf^1
+      const bool load_next2 = t + 2 < KT;    // load t+2 into regs
+      // ph1
+      READ_B13(Bt);
+      READ_A13(At, 0);
+      if (stage_next) WRITE_REGS(aregs, Aw);
+      __builtin_amdgcn_s_barrier();
+      PHASE_MFMA13(0);
+      // ph2
+      READ_A13(At, 1);
+      if (stage_next) WRITE_REGS(bregs, Bw);
+      if (load_next2) LOAD_REGS(aregs, A, K, bm, M, (t + 2) * BK);
+      __builtin_amdgcn_s_barrier();
+      PHASE_MFMA13(1);
+      // ph3
+      READ_A13(At, 2);
+      if (load_next2) LOAD_REGS(bregs, B, K, bn, N, (t + 2) * BK);
+      __builtin_amdgcn_s_barrier();
+      PHASE_MFMA13(2);
+      // ph4
+      READ_A13(At, 3);
+      __builtin_amdgcn_s_barrier();
+      PHASE_MFMA13(3);
+    }
+
+    // persistent: issue the NEXT output tile's prologue loads before the
+    // epilogue stores so their HBM latency hides under the writeback.
+    if (PERSIST && rep + 1 < tiles) {
+      int nxt = (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (nxt < nwg) {
+        int o2 = nxt;
+        if (do_remap) {
+          int q = nwg >> 3, r = nwg & 7;
+          int xcd = o2 & 7, lid = o2 >> 3;
+          o2 = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+        }
+        const long bm2 = (long)(o2 / nbx) * BM;
+        const long bn2 = (long)(o2 % nbx) * BN;
+        LOAD_REGS(aregs, A, K, bm2, M, 0);
+        LOAD_REGS(bregs, B, K, bn2, N, 0);
+      }
+    }
+
+    // epilogue (same C/D map as production)
+    const long crow_base = bm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = bn + waveN * WN + (lane & 15);
+    const bool interior = (bm + BM <= M) && (bn + BN <= N);
+    if (interior) {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (col >= N) continue;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (row >= M) continue;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    }
+    if (PERSIST && rep + 1 < tiles) __syncthreads();  // LDS reuse fence
+  }
+#undef A13
+#undef B13
+#undef PHASE_MFMA13
+#undef READ_A13
+#undef READ_B13
+}
+
+}  // namespace
+
+// persist: 0 = one tile per WG (grid = nwg), 1 = persistent (grid =
+// min(nwg, 8*wgs_per_xcd*... caller passes grid via the env knob below).
+extern "C" int cc_gemm_v13(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int persist,
+                           int remap, unsigned long long stream) {
+  if (K % BK != 0 || K / BK < 2) return -2;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg;
+  if (persist) grid = nwg < 256 ? nwg : 256;  // 1 WG/CU persistent fleet
+  if (persist)
+    hipLaunchKernelGGL((k_gemm_v13<1>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  else
+    hipLaunchKernelGGL((k_gemm_v13<0>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// ---- V15: production glds 8-phase schedule + persistent multi-tile ----
+// v13 showed register staging costs more than the glds path's vmcnt(0)
+// drain (704 vs 791 at patch64), but persistence + remap wins at the
+// skinny-N ViT shapes (fc1 784 vs 705).  v15 keeps the production
+// KTILE2 glds schedule verbatim and adds the persistent outer loop,
+// issuing the NEXT tile's 12-glds prologue BEFORE the epilogue stores
+// so its HBM latency hides under the C writeback.
+namespace {
+
+__device__ __forceinline__ void stage_half15(const __bf16* __restrict__ src,
+                                             long ld, long row0,
+                                             long row_limit, long k0,
+                                             __bf16* lds_rowbase, int lane) {
+  const int lrow8 = lane >> 3;
+  const int slot = lane & 7;
+  const int gk16 = slot ^ lrow8;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    long grow = row0 + j * 8 + lrow8;
+    grow = grow < 0 ? 0 : (grow >= row_limit ? row_limit - 1 : grow);
+    const __bf16* gptr = src + grow * ld + k0 + (long)gk16 * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gptr,
+        (__attribute__((address_space(3))) unsigned int*)(lds_rowbase +
+                                                          j * 8 * BK),
+        16, 0, 0);
+  }
+}
+
+template <int PERSIST>
+__global__ __launch_bounds__(512, 1) void k_gemm_v15(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // 128 KiB
+#define A15(b) (lds + (b) * (BM * BK))
+#define B15(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long KT = K / BK;  // caller guarantees even, >= 4
+  const long srow = wid * 16;
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2];
+
+#define STAGE_A15(bmv, t, h)                                                \
+  stage_half15(A, K, (bmv) + (h) * 128 + srow, M, (t) * BK,                 \
+               A15((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B15(bnv, t, h)                                                \
+  stage_half15(B, K, (bnv) + (h) * 128 + srow, N, (t) * BK,                 \
+               B15((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+  // tile index for rep r of this WG under the three order modes:
+  //   0: orig = bid + r*grid                  (stride)
+  //   1: xcd-remap(orig) of mode 0            (stride + L2 affinity)
+  //   2: orig = chunk_start(bid) + r          (contiguous row-major chunk:
+  //      the A row band stays hot in L2 across ~nbx consecutive reps)
+  int chunk_start = 0, chunk_len = 1;
+  if (PERSIST && do_remap == 2) {
+    int q = nwg / (int)gridDim.x, r = nwg % (int)gridDim.x;
+    int b = (int)blockIdx.x;
+    chunk_len = q + (b < r ? 1 : 0);
+    chunk_start = b * q + (b < r ? b : r);
+  }
+  const int tiles =
+      PERSIST ? (do_remap == 2 ? chunk_len
+                               : (nwg + (int)gridDim.x - 1) / (int)gridDim.x)
+              : 1;
+  long bm = 0, bn = 0;
+  {
+    int orig = do_remap == 2 ? chunk_start : (int)blockIdx.x;
+    if (do_remap == 1) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    bm = (long)(orig / nbx) * BM;
+    bn = (long)(orig % nbx) * BN;
+  }
+  // first prologue (full-latency; later ones hide under epilogues)
+  STAGE_A15(bm, 0, 0);
+  STAGE_A15(bm, 0, 1);
+  STAGE_B15(bn, 0, 0);
+  STAGE_B15(bn, 0, 1);
+  STAGE_B15(bn, 1, 0);
+  STAGE_B15(bn, 1, 1);
+
+  for (int rep = 0; rep < tiles; rep++) {
+    if (PERSIST && do_remap != 2 &&
+        (int)blockIdx.x + rep * (int)gridDim.x >= nwg)
+      break;
+    f32x4 acc[MFR][NFR] = {};
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+#define PHASE_MFMA15(q)                                                     \
+  do {                                                                      \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define READ_A15(At, q)                                                     \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =             \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B15(Bt)                                                        \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+#define KTILE15(t)                                                          \
+  do {                                                                      \
+    const __bf16* At = A15((t) & 1);                                        \
+    const __bf16* Bt = B15((t) & 1);                                        \
+    READ_B15(Bt);                                                           \
+    READ_A15(At, 0);                                                        \
+    if ((t) + 1 < KT) STAGE_A15(bm, (t) + 1, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA15(0);                                                        \
+    READ_A15(At, 1);                                                        \
+    if ((t) + 1 < KT) STAGE_A15(bm, (t) + 1, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA15(1);                                                        \
+    READ_A15(At, 2);                                                        \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    if ((t) + 2 < KT) STAGE_B15(bn, (t) + 2, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA15(2);                                                        \
+    READ_A15(At, 3);                                                        \
+    if ((t) + 2 < KT) STAGE_B15(bn, (t) + 2, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA15(3);                                                        \
+  } while (0)
+
+    for (long it = 0; it < KT / 2; ++it) {
+      KTILE15(2 * it);
+      KTILE15(2 * it + 1);
+    }
+
+    const long ebm = bm, ebn = bn;
+    // advance to next tile and issue its prologue BEFORE the epilogue:
+    // the LDS buffers are fully consumed (every wave drained its reads
+    // via lgkmcnt(0) before its last barrier), but another wave may
+    // still be inside the last phase -> barrier first.
+    if (PERSIST && rep + 1 < tiles &&
+        (do_remap == 2 ||
+         (int)blockIdx.x + (rep + 1) * (int)gridDim.x < nwg)) {
+      __builtin_amdgcn_s_barrier();
+      int orig = do_remap == 2
+                     ? chunk_start + rep + 1
+                     : (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (do_remap == 1) {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = orig & 7, lid = orig >> 3;
+        orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+      }
+      bm = (long)(orig / nbx) * BM;
+      bn = (long)(orig % nbx) * BN;
+      STAGE_A15(bm, 0, 0);
+      STAGE_A15(bm, 0, 1);
+      STAGE_B15(bn, 0, 0);
+      STAGE_B15(bn, 0, 1);
+      STAGE_B15(bn, 1, 0);
+      STAGE_B15(bn, 1, 1);
+    }
+
+    const long crow_base = ebm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN + (lane & 15);
+    const bool interior = (ebm + BM <= M) && (ebn + BN <= N);
+    if (interior) {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (col >= N) continue;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (row >= M) continue;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    }
+  }
+#undef A15
+#undef B15
+#undef STAGE_A15
+#undef STAGE_B15
+#undef PHASE_MFMA15
+#undef READ_A15
+#undef READ_B15
+#undef KTILE15
+}
+
+}  // namespace
+
+extern "C" int cc_gemm_v15(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int persist,
+                           int remap, unsigned long long stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4) return -2;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg;
+  if (persist) grid = nwg < 256 ? nwg : 256;
+  if (persist)
+    hipLaunchKernelGGL((k_gemm_v15<1>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  else
+    hipLaunchKernelGGL((k_gemm_v15<0>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
+
+// ---- V16: v15 + per-XCD atomic tile queues ----
+// v15's static chunking leaves a straggler tail when nwg/grid has a big
+// fractional part (out64: 789/256 = 3.08 -> 21 WGs run a 4th tile while
+// 235 CUs idle).  v16 pops tile indices from one atomic counter per XCD
+// over a contiguous row-major range: dynamic balance inside each XCD's
+// 32 CUs + the contiguous window keeps A row bands hot in that XCD's L2.
+// Results stay bitwise deterministic (each tile's C block is a pure
+// function of A/B; only WHO computes it varies run to run).
+namespace {
+
+template <int MODE>  // 0: global queue; 1: per-XCD ranges
+__global__ __launch_bounds__(512, 1) void k_gemm_v16(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int* __restrict__ ctrs) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // 128 KiB
+  __shared__ int next_tile_s;
+#define A16(b) (lds + (b) * (BM * BK))
+#define B16(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long KT = K / BK;
+  const long srow = wid * 16;
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2];
+
+  // this WG's queue: XCD id from the dispatch round-robin (T1)
+  const int xcd = MODE ? ((int)blockIdx.x & 7) : 0;
+  int rstart = 0, rlen = nwg;
+  if (MODE) {
+    const int q = nwg >> 3, r = nwg & 7;
+    rlen = q + (xcd < r ? 1 : 0);
+    rstart = xcd * q + (xcd < r ? xcd : r);
+  }
+
+#define STAGE_A16(bmv, t, h)                                                \
+  stage_half15(A, K, (bmv) + (h) * 128 + srow, M, (t) * BK,                 \
+               A16((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B16(bnv, t, h)                                                \
+  stage_half15(B, K, (bnv) + (h) * 128 + srow, N, (t) * BK,                 \
+               B16((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+#define POP_TILE(dst)                                                       \
+  do {                                                                      \
+    if (tid == 0) next_tile_s = atomicAdd(ctrs + xcd, 1);                   \
+    __syncthreads();                                                        \
+    dst = next_tile_s;                                                      \
+  } while (0)
+
+  int idx;
+  POP_TILE(idx);
+  if (idx >= rlen) return;
+  long bm = (long)((rstart + idx) / nbx) * BM;
+  long bn = (long)((rstart + idx) % nbx) * BN;
+  STAGE_A16(bm, 0, 0);
+  STAGE_A16(bm, 0, 1);
+  STAGE_B16(bn, 0, 0);
+  STAGE_B16(bn, 0, 1);
+  STAGE_B16(bn, 1, 0);
+  STAGE_B16(bn, 1, 1);
+
+  while (true) {
+    f32x4 acc[MFR][NFR] = {};
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+#define PHASE_MFMA16(q)                                                     \
+  do {                                                                      \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define READ_A16(At, q)                                                     \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =             \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B16(Bt)                                                        \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+#define KTILE16(t)                                                          \
+  do {                                                                      \
+    const __bf16* At = A16((t) & 1);                                        \
+    const __bf16* Bt = B16((t) & 1);                                        \
+    READ_B16(Bt);                                                           \
+    READ_A16(At, 0);                                                        \
+    if ((t) + 1 < KT) STAGE_A16(bm, (t) + 1, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA16(0);                                                        \
+    READ_A16(At, 1);                                                        \
+    if ((t) + 1 < KT) STAGE_A16(bm, (t) + 1, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA16(1);                                                        \
+    READ_A16(At, 2);                                                        \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    if ((t) + 2 < KT) STAGE_B16(bn, (t) + 2, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA16(2);                                                        \
+    READ_A16(At, 3);                                                        \
+    if ((t) + 2 < KT) STAGE_B16(bn, (t) + 2, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA16(3);                                                        \
+  } while (0)
+
+    for (long it = 0; it < KT / 2; ++it) {
+      KTILE16(2 * it);
+      KTILE16(2 * it + 1);
+    }
+
+    const long ebm = bm, ebn = bn;
+    // pop + stage the next tile before the epilogue (latency hides
+    // under the C stores).  POP_TILE carries its own __syncthreads,
+    // which also fences the LDS reuse.
+    POP_TILE(idx);
+    const bool more = idx < rlen;
+    if (more) {
+      bm = (long)((rstart + idx) / nbx) * BM;
+      bn = (long)((rstart + idx) % nbx) * BN;
+      STAGE_A16(bm, 0, 0);
+      STAGE_A16(bm, 0, 1);
+      STAGE_B16(bn, 0, 0);
+      STAGE_B16(bn, 0, 1);
+      STAGE_B16(bn, 1, 0);
+      STAGE_B16(bn, 1, 1);
+    }
+
+    const long crow_base = ebm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN + (lane & 15);
+    const bool interior = (ebm + BM <= M) && (ebn + BN <= N);
+    if (interior) {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (col >= N) continue;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (row >= M) continue;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    }
+    if (!more) return;
+  }
+#undef A16
+#undef B16
+#undef STAGE_A16
+#undef STAGE_B16
+#undef PHASE_MFMA16
+#undef READ_A16
+#undef READ_B16
+#undef KTILE16
+#undef POP_TILE
+}
+
+}  // namespace
+
+extern "C" int cc_gemm_v16(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int mode,
+                           unsigned long long stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4) return -2;
+  static int* ctrs = nullptr;
+  if (!ctrs && hipMalloc(&ctrs, 8 * sizeof(int)) != hipSuccess) return -3;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg < 256 ? nwg : 256;
+  hipMemsetAsync(ctrs, 0, 8 * sizeof(int), (hipStream_t)stream);
+  if (mode)
+    hipLaunchKernelGGL((k_gemm_v16<1>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       ctrs);
+  else
+    hipLaunchKernelGGL((k_gemm_v16<0>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       ctrs);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
