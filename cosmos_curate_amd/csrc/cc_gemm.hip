@@ -65,6 +65,20 @@ __device__ __forceinline__ unsigned short f32_to_bf16_rne(float v) {
   return (unsigned short)((cv.u + 0x7fffu + lsb) >> 16);
 }
 
+// fused activations (quick-gelu / SigLIP tanh-gelu; see epilogue notes)
+template <int ACT>
+__device__ __forceinline__ float cc_act(float v) {
+  if constexpr (ACT == 1)
+    return v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
+  if constexpr (ACT == 2) {
+    // tanh(z) = 1 - 2/(e^{2z}+1): __expf+rcp beats libm tanhf
+    float z = 0.7978845608028654f * (v + 0.044715f * v * v * v);
+    float t = 1.0f - 2.0f * __builtin_amdgcn_rcpf(__expf(2.0f * z) + 1.0f);
+    return 0.5f * v * (1.0f + t);
+  }
+  return v;
+}
+
 // stage one 32-row slice of a [rows x BK] bf16 tile into LDS via glds,
 // with the k16 ^ (row&7) source swizzle.  lds_base = this wave's slice.
 __device__ __forceinline__ void stage_slice(const __bf16* __restrict__ src,
@@ -426,7 +440,18 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int do_remap) {
+    int c_is_bf16, int nbx, int nwg, int order_mode) {
+  // Round-2 structure: PERSISTENT multi-tile (grid = min(nwg, 256), one
+  // WG per CU) around the round-1 8-phase glds schedule.  Each WG walks
+  // tiles bid, bid+grid, ... and issues the NEXT tile's 12-glds
+  // prologue BEFORE the epilogue stores, so prologue HBM latency hides
+  // under the C writeback instead of stalling a fresh WG.  Measured
+  // +18-27% at every batch-64 ViT shape and +13% at square8k
+  // (profiles/r02_gemm_persist.log: fc1 711->874, fc2 795->968, patch
+  // 814->964, qkv 724->887 TF within one box).  order_mode: 0 = plain
+  // stride; 1 = stride + bijective XCD remap (skinny-N grids, L2 row
+  // sharing).  Sync structure per K-tile is UNCHANGED from the
+  // race-screened v11 (vmcnt(0) publish at ph3, leading barriers).
   __shared__ __bf16 lds[2 * (BM2 + BN2) * BK];  // 128 KiB
 #define A2T(b) (lds + (b) * (BM2 * BK))
 #define B2T(b) (lds + 2 * (BM2 * BK) + (b) * (BN2 * BK))
@@ -434,18 +459,12 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int waveM = wid >> 2, waveN = wid & 3;
-  int orig = blockIdx.x;
-  if (do_remap) {
-    int q = nwg >> 3, r = nwg & 7;
-    int xcd = orig & 7, lid = orig >> 3;
-    orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
-  }
-  const long bm = (long)(orig / nbx) * BM2;
-  const long bn = (long)(orig % nbx) * BN2;
-
-  f32x4 acc[MFR2][NFR2] = {};
   const long KT = K / BK;  // launcher guarantees even, >= 4
   const long srow = wid * 16;
+  const int arow_base = waveM * WM2 + (lane & 15);
+  const int brow_base = waveN * WN2 + (lane & 15);
+  bf16x8 bfragT[NFR2][2];  // held for the whole K-tile
+  bf16x8 afrag[2][2];      // one quadrant: 2 M-frags x 2 k-steps
 
 #define STAGE_A2(t, h)                                                      \
   stage_half(A, K, bm + (h) * 128 + srow, M, (t) * BK,                      \
@@ -454,21 +473,32 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
   stage_half(B, K, bn + (h) * 128 + srow, N, (t) * BK,                      \
              B2T((t) & 1) + ((h) * 128 + srow) * BK, lane)
 
-  // prologue: A(0) h0,h1, B(0) h0,h1, B(1) h0,h1 (12 instrs); wait for
-  // tile 0's 8 (the 4 newest = B(1) stay in flight), publish.
+  const int tiles = (nwg + (int)gridDim.x - 1) / (int)gridDim.x;
+  long bm, bn;
+  {
+    int orig = (int)blockIdx.x;
+    if (order_mode == 1) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    bm = (long)(orig / nbx) * BM2;
+    bn = (long)(orig % nbx) * BN2;
+  }
+  // first prologue: A(0) h0,h1, B(0) h0,h1, B(1) h0,h1 (12 glds); later
+  // tiles' prologues issue under the previous epilogue.
   STAGE_A2(0, 0);
   STAGE_A2(0, 1);
   STAGE_B2(0, 0);
   STAGE_B2(0, 1);
   STAGE_B2(1, 0);
   STAGE_B2(1, 1);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // full drain: counted
-  __builtin_amdgcn_s_barrier();                     // waits are unsound (above)
 
-  const int arow_base = waveM * WM2 + (lane & 15);
-  const int brow_base = waveN * WN2 + (lane & 15);
-  bf16x8 bfragT[NFR2][2];  // held for the whole K-tile
-  bf16x8 afrag[2][2];      // one quadrant: 2 M-frags x 2 k-steps
+  for (int rep = 0; rep < tiles; rep++) {
+    if ((int)blockIdx.x + rep * (int)gridDim.x >= nwg) break;
+    f32x4 acc[MFR2][NFR2] = {};
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // prologue drain
+    __builtin_amdgcn_s_barrier();                     // (counted waits unsound: det2)
 
 #define PHASE_MFMA2(q)                                                      \
   do {                                                                      \
@@ -539,77 +569,118 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     PHASE_MFMA2(3);                                                         \
   } while (0)
 
-  for (long it = 0; it < KT / 2; ++it) {
-    KTILE2(2 * it);
-    KTILE2(2 * it + 1);
-  }
+    for (long it = 0; it < KT / 2; ++it) {
+      KTILE2(2 * it);
+      KTILE2(2 * it + 1);
+    }
 
-  // epilogue: same C/D map as the 128² body, MFR2 x NFR2 fragments
-  const long crow_base = bm + waveM * WM2 + 4 * (lane >> 4);
-  const long ccol_base = bn + waveN * WN2 + (lane & 15);
-  const bool interior = (bm + BM2 <= M) && (bn + BN2 <= N);
-  if (interior) {
+    const long ebm = bm, ebn = bn;
+    // barrier before the epilogue: the bf16 epilogue below stages C
+    // through A-buf1, whose rows 96-127/224-255 other waves were still
+    // reading in the last phase; the next tile's prologue (further
+    // below) overwrites A-buf0/B-bufs with the same hazard.
+    __builtin_amdgcn_s_barrier();
+    // advance + issue the next tile's prologue BEFORE the epilogue so
+    // its 12-glds HBM latency hides under the C writeback.
+    if (rep + 1 < tiles &&
+        (int)blockIdx.x + (rep + 1) * (int)gridDim.x < nwg) {
+      int orig = (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (order_mode == 1) {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = orig & 7, lid = orig >> 3;
+        orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+      }
+      bm = (long)(orig / nbx) * BM2;
+      bn = (long)(orig % nbx) * BN2;
+      STAGE_A2(0, 0);
+      STAGE_A2(0, 1);
+      STAGE_B2(0, 0);
+      STAGE_B2(0, 1);
+      STAGE_B2(1, 0);
+      STAGE_B2(1, 1);
+    }
+
+    // ---- epilogue ----
+    // bf16 output (all tower-internal GEMMs): LDS-staged + coalesced.
+    // The fragment C/D map scatters 2-byte lanes (col = lane&15, row =
+    // 4*(lane>>4)+r): storing it directly costs one 2 B store per
+    // element and the same pattern for the fused residual loads — the
+    // dominant cost at the thin ViT shapes once the K-loop is fed
+    // (attn_out 429 vs ~790 TF bare, profiles/r02_persist_ab.log).
+    // Instead: 4 passes of 2 M-fragments stage bf16(act(acc+bias))
+    // through this wave's free 4 KB slice of A-buf1 (last read before
+    // the barrier above; the next prologue writes only A-buf0/B-bufs),
+    // then read back row-contiguous 16 B chunks, fuse the residual add
+    // in f32, and store full b128s (8 lanes cover 128 contiguous
+    // bytes of a C row).  Residual numerics: the pre-residual value is
+    // rounded to bf16 once before the f32 add — one extra rounding vs
+    // the scalar path, inside the embedding-cosine contract.
+    const long crow_base = ebm + waveM * WM2 + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN2 + (lane & 15);
+    if (c_is_bf16) {
+      __bf16* slice = A2T(1) + wid * 2048;  // 4 KB per wave
+      const long wrow0 = ebm + waveM * WM2;
+      const long wcol0 = ebn + waveN * WN2;
 #pragma unroll
-    for (int m = 0; m < MFR2; m++) {
+      for (int p = 0; p < 4; p++) {
 #pragma unroll
-      for (int n = 0; n < NFR2; n++) {
-        const long col = ccol_base + n * FRAG;
-        float bval = 0.0f;
-        if constexpr (HAS_BIAS) bval = bias[col];
-        float rv[4];
-        if constexpr (HAS_RES) {
+        for (int mm = 0; mm < 2; mm++) {
+          const int m = 2 * p + mm;
 #pragma unroll
-          for (int r = 0; r < 4; r++)
-            rv[r] = (float)residual[(crow_base + m * FRAG + r) * N + col];
-        }
+          for (int n = 0; n < NFR2; n++) {
+            const long col = ccol_base + n * FRAG;
+            float bval = 0.0f;
+            if constexpr (HAS_BIAS) bval = (col < N) ? bias[col] : 0.0f;
+            const int lrow_b = mm * 16 + 4 * (lane >> 4);
+            const int lcol = n * FRAG + (lane & 15);
 #pragma unroll
-        for (int r = 0; r < 4; r++) {
-          const long row = crow_base + m * FRAG + r;
-          float v = acc[m][n][r] + bval;
-          if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
-          if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
-            // tanh(z) = 1 - 2/(e^{2z}+1): __expf+rcp beats libm tanhf
-            // (~80 us/launch at the SigLIP fc1 shape, r01_siglip_prof)
-            float z = 0.7978845608028654f * (v + 0.044715f * v * v * v);
-            float t = 1.0f - 2.0f * __builtin_amdgcn_rcpf(__expf(2.0f * z) + 1.0f);
-            v = 0.5f * v * (1.0f + t);
+            for (int r = 0; r < 4; r++) {
+              float v = cc_act<ACT>(acc[m][n][r] + bval);
+              *(unsigned short*)(slice + (lrow_b + r) * 64 + lcol) =
+                  f32_to_bf16_rne(v);
+            }
           }
-          if constexpr (HAS_RES) v += rv[r];
-          if (c_is_bf16)
-            ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
-          else
-            ((float*)C)[row * N + col] = v;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+          const int elem = i * 512 + lane * 8;
+          const int lrow = elem >> 6;          // /64
+          const int lcol8 = elem & 63;         // 8-elem aligned
+          const long grow = wrow0 + p * 32 + lrow;
+          const long gcol = wcol0 + lcol8;
+          if (grow >= M || gcol >= N) continue;  // N % 8 == 0 always
+          bf16x8 v8 = *(const bf16x8*)(slice + elem);
+          if constexpr (HAS_RES) {
+            const bf16x8 r8 =
+                *(const bf16x8*)(residual + grow * N + gcol);
+#pragma unroll
+            for (int e = 0; e < 8; e++)
+              v8[e] = (__bf16)((float)v8[e] + (float)r8[e]);
+          }
+          *(bf16x8*)((__bf16*)C + grow * N + gcol) = v8;
         }
       }
-    }
-    return;
-  }
+    } else {
+      // f32 output (visual projection / probes): scalar store path
+      const bool interior = (ebm + BM2 <= M) && (ebn + BN2 <= N);
 #pragma unroll
-  for (int m = 0; m < MFR2; m++) {
+      for (int m = 0; m < MFR2; m++) {
 #pragma unroll
-    for (int n = 0; n < NFR2; n++) {
-      const long col = ccol_base + n * FRAG;
-      if (col >= N) continue;
-      float bval = 0.0f;
-      if constexpr (HAS_BIAS) bval = bias[col];
+        for (int n = 0; n < NFR2; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (!interior && col >= N) continue;
+          float bval = 0.0f;
+          if constexpr (HAS_BIAS) bval = bias[col];
 #pragma unroll
-      for (int r = 0; r < 4; r++) {
-        const long row = crow_base + m * FRAG + r;
-        if (row >= M) continue;
-        float v = acc[m][n][r] + bval;
-        if (ACT == 1) v = v * __builtin_amdgcn_rcpf(1.0f + __expf(-1.702f * v));
-        if (ACT == 2) {  // tanh-gelu (SigLIP gelu_pytorch_tanh)
-          // tanh(z) = 1 - 2/(e^{2z}+1): __expf+rcp beats libm tanhf
-          // (~80 us/launch at the SigLIP fc1 shape, r01_siglip_prof)
-          float z = 0.7978845608028654f * (v + 0.044715f * v * v * v);
-          float t = 1.0f - 2.0f * __builtin_amdgcn_rcpf(__expf(2.0f * z) + 1.0f);
-          v = 0.5f * v * (1.0f + t);
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (!interior && row >= M) continue;
+            float v = cc_act<ACT>(acc[m][n][r] + bval);
+            if constexpr (HAS_RES) v += (float)residual[row * N + col];
+            ((float*)C)[row * N + col] = v;
+          }
         }
-        if constexpr (HAS_RES) v += (float)residual[row * N + col];
-        if (c_is_bf16)
-          ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
-        else
-          ((float*)C)[row * N + col] = v;
       }
     }
   }
@@ -661,24 +732,39 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
     return e ? atoi(e) : -1;
   }();
   const bool t256_ok = (K % 128 == 0) && (K >= 256) && M > BM2 / 2;
+  // round-2: the t256 body is persistent (grid = min(nwg, 256)); it now
+  // also wins the N=768 && K=768 shape (out64 720 -> 796 TF,
+  // profiles/r02_gemm_persist.log), so the only shapes kept on the 128
+  // body are those whose 256-tile grid would underfill the persistent
+  // fleet (nwg256 < 256, e.g. the tiny visual-projection GEMM).
+  const int nbx2 = (int)((N + BN2 - 1) / BN2);
+  const int nby2 = (int)((M + BM2 - 1) / BM2);
+  const int nwg2 = nbx2 * nby2;
   const bool t256 =
       !wide && t256_ok &&
-      (tile_env >= 0 ? tile_env != 0 : (K >= 1536 || N >= 1536));
+      (tile_env >= 0 ? tile_env != 0
+                     : (K >= 1536 || N >= 1536 || nwg2 >= 256));
   const bool hb = bias != nullptr;
   const bool hr = residual != nullptr;
   // XCD remap only for outputs that spill L3 (measured negative on the
   // L2-resident ViT shapes, profiles/r01_opt3)
   int remap = ((M * N) > (48LL << 20)) ? 1 : 0;
   if (t256) {
-    nbx = (int)((N + BN2 - 1) / BN2);
-    nby = (int)((M + BM2 - 1) / BM2);
-    nwg = nbx * nby;
-    grid = dim3(nwg);
+    nbx = nbx2;
+    nby = nby2;
+    nwg = nwg2;
+    // persistent fleet: 1 WG/CU; CC_GEMM_PERSIST=0 restores grid = nwg
+    // (one tile per WG) for A/B experiments.
+    static const int persist_env = [] {
+      const char* e = getenv("CC_GEMM_PERSIST");
+      return e ? atoi(e) : 1;
+    }();
+    grid = dim3(persist_env && nwg > 256 ? 256 : nwg);
     block = dim3(512);
-    // for the 256² body, row-band L2 sharing via the remap pays only on
-    // skinny-N grids (qkv64 +15%, fc1_64 +12%); on wide grids it costs
-    // (square8k 962 vs 1168 TF) — profiles/r01_t256_fixed.log
-    if (nbx > 12) remap = 0;
+    // stride + XCD remap order pays on skinny-N grids (row-band L2
+    // sharing: qkv/fc1/fc2/patch +9-13%); on wide grids plain stride
+    // wins (square8k 1295 vs 944 — profiles/r02_gemm_persist.log)
+    remap = nbx <= 16 ? 1 : 0;
   }
 #define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
   hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \

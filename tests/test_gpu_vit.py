@@ -463,7 +463,7 @@ def test_clip_non224_input_resized_on_device(lib):
     # CPU reference of the same chain: shorter side -> 224 bicubic, center crop
     scale = 224 / 240
     rh, rw = max(224, round(240 * scale)), max(224, round(320 * scale))
-    resized = resize_bicubic_u8(frames, rh, rw)
+    resized = np.stack([resize_bicubic_u8(f, rh, rw) for f in frames])
     top, left = (rh - 224) // 2, (rw - 224) // 2
     cropped = resized[:, top:top + 224, left:left + 224, :]
     ref = oracle_vit.build_reference_clip_vision(make_clip_vit_b32_weights())
