@@ -42,6 +42,15 @@ def _declare(lib: ctypes.CDLL) -> None:
         c.POINTER(c.c_size_t)]
     lib.cc_buffer_free.argtypes = [c.c_void_p]
 
+    lib.cc_decode_session_create.argtypes = [
+        c.c_int, c.c_int32, c.POINTER(c.c_void_p)]
+    lib.cc_decode_submit.argtypes = [
+        c.c_void_p, c.c_char_p, c.c_size_t, c.c_int64]
+    lib.cc_decode_map_frames.argtypes = [
+        c.c_void_p, c.c_void_p, c.c_size_t, c.POINTER(c.c_size_t)]
+    lib.cc_decode_recycle.argtypes = [c.c_void_p]
+    lib.cc_decode_destroy.argtypes = [c.c_void_p]
+
     lib.cc_malloc.argtypes = [c.POINTER(c.c_void_p), c.c_size_t]
     lib.cc_free.argtypes = [c.c_void_p]
     lib.cc_memcpy_h2d.argtypes = [c.c_void_p, c.c_void_p, c.c_size_t, c.c_uint64]
@@ -142,6 +151,64 @@ class VideoInfo(ctypes.Structure):
         ("duration_s", ctypes.c_double),
         ("avg_fps", ctypes.c_double),
     ]
+
+
+class Nv12Frame(ctypes.Structure):
+    """cc_nv12_frame: a mapped VCN NV12 surface (device pointers)."""
+
+    _fields_ = [
+        ("y", ctypes.c_void_p),
+        ("uv", ctypes.c_void_p),
+        ("pitch", ctypes.c_size_t),
+        ("pts", ctypes.c_int64),
+        ("width", ctypes.c_uint32),
+        ("height", ctypes.c_uint32),
+    ]
+
+
+class DecodeSession:
+    """RAII wrapper over cc_decode_* (the VCN/rocDecode seam,
+    nvcodec_utils.py:199-313 NvVideoDecoder counterpart).
+
+    Raises on construction with CC_ERR_NO_ROCDECODE when librocdecode is
+    absent — loud, no CPU fallback.  Mapped surfaces stay valid until
+    ``recycle()``; consume (launch + synchronize) first.
+    """
+
+    def __init__(self, device: int = 0, codec: int = 0) -> None:
+        lib = load()
+        h = ctypes.c_void_p()
+        check(lib.cc_decode_session_create(device, codec, ctypes.byref(h)))
+        self._lib = lib
+        self._h: ctypes.c_void_p | None = h
+
+    def submit(self, packet: bytes | None, pts: int = 0) -> None:
+        """One AnnexB access unit; None flushes (end of stream)."""
+        if packet is None:
+            check(self._lib.cc_decode_submit(self._h, None, 0, 0))
+        else:
+            check(self._lib.cc_decode_submit(self._h, packet, len(packet), pts))
+
+    def map_frames(self, cap: int = 32) -> list[Nv12Frame]:
+        arr = (Nv12Frame * cap)()
+        n = ctypes.c_size_t()
+        check(self._lib.cc_decode_map_frames(
+            self._h, ctypes.byref(arr), cap, ctypes.byref(n)))
+        return [arr[i] for i in range(n.value)]
+
+    def recycle(self) -> None:
+        check(self._lib.cc_decode_recycle(self._h))
+
+    def close(self) -> None:
+        if self._h is not None:
+            self._lib.cc_decode_destroy(self._h)
+            self._h = None
+
+    def __del__(self) -> None:  # pragma: no cover - GC timing
+        try:
+            self.close()
+        except Exception:
+            pass
 
 
 class Demuxer:

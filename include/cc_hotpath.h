@@ -96,13 +96,22 @@ int cc_stream_sync(uint64_t stream);
  * (nvcodec_utils.py:199-313).  Fails with CC_ERR_NO_ROCDECODE when
  * librocdecode.so is absent (this image ships none; the ABI is the seam). */
 typedef struct cc_decode cc_decode_t;
+/* codec: 0 = h264, 1 = hevc (cc_video_info.codec values). */
 int  cc_decode_session_create(int device, int32_t codec, cc_decode_t** out);
+/* One AnnexB access unit (cc_demux_packet output) per call; pkt == NULL
+ * flushes the stream (end-of-stream). */
 int  cc_decode_submit(cc_decode_t* s, const uint8_t* pkt, size_t size, int64_t pts);
-/* Mapped NV12 surfaces land as {y_ptr, uv_ptr, pitch} triples. */
+/* Mapped NV12 surfaces land as {y_ptr, uv_ptr, pitch} triples (HIP device
+ * pointers into the decoder's surface pool).  Surfaces stay valid until
+ * cc_decode_recycle; consume them (launch + synchronize) first. */
 typedef struct cc_nv12_frame {
   void* y; void* uv; size_t pitch; int64_t pts; uint32_t width, height;
 } cc_nv12_frame;
 int  cc_decode_map_frames(cc_decode_t* s, cc_nv12_frame* out, size_t cap, size_t* n);
+/* Return every surface handed out by map_frames to the decoder pool
+ * (rocDecParserMarkFrameForReuse).  Call after the consuming kernels
+ * complete. */
+int  cc_decode_recycle(cc_decode_t* s);
 void cc_decode_destroy(cc_decode_t* s);
 
 /* ---- fused pixel kernels (hand-written HIP, gfx950) -----------------
