@@ -221,19 +221,29 @@ def chunk_tasks(
     return out
 
 
-def _slice_raw_clip(raw: bytes, span: tuple[float, float]) -> bytes:
-    """Frame-exact raw-NV12 span slice: frames with i/fps in [start, end)."""
+def _slice_raw_clip(raw: bytes | np.ndarray, span: tuple[float, float]) -> np.ndarray:
+    """Frame-exact raw-NV12 span slice: frames with i/fps in [start, end).
+
+    Zero-copy when the span covers the whole source; otherwise one
+    header+body concatenation (frames are contiguous in the payload), no
+    per-frame gather — raw payloads are ~93 MB per second of video, so
+    every avoided pass shows up in the driver-level throughput
+    (profiles/r02_driver_bench.log vs r01's 5.3 s transcode stage).
+    """
     n, h, w, fps = raw_backend.parse_header(raw)
     first = int(math.ceil(span[0] * fps - 1e-6))
     last = int(math.ceil(span[1] * fps - 1e-6))  # exclusive
     first = max(0, min(first, n))
     last = max(first, min(last, n))
+    buf = np.frombuffer(raw, dtype=np.uint8)
     if first == 0 and last == n:
-        return raw  # span covers the whole source: zero-copy payload
-    idx = np.arange(first, last, dtype=np.int32)
-    ys, uvs = raw_backend.frame_planes(raw, idx)
+        return buf  # span covers the whole source: zero-copy payload
+    fsz = h * w + (h // 2) * w
     num = int(round(fps))
-    return raw_backend.encode_raw_nv12(ys, uvs, num, 1)
+    hdr = raw_backend.pack_header(last - first, h, w, num, 1)
+    body = buf[raw_backend.HEADER_SIZE + first * fsz:
+               raw_backend.HEADER_SIZE + last * fsz]
+    return np.concatenate([np.frombuffer(hdr, dtype=np.uint8), body])
 
 
 def _remux_mp4_clip(raw: bytes, span: tuple[float, float]) -> bytes:
@@ -293,12 +303,17 @@ class ClipTranscodingStage(CuratorStage):
         if data is None:
             video.errors["transcode"] = "no encoded data"
             return
-        raw = bytes(data) if not isinstance(data, bytes) else data
-        is_raw = raw_backend.is_raw_nv12(raw)
+        is_raw = raw_backend.is_raw_nv12(data)
         video.was_remuxed = not is_raw  # mp4 spans get new containers
+        # raw payloads stay numpy end to end (no bytes() copy of the
+        # ~GB source); mp4 payloads (real-codec sized) go through the
+        # demuxer, which wants bytes
+        mp4_bytes = None if is_raw else (
+            data if isinstance(data, bytes) else bytes(data))
         for clip in video.clips:
             try:
-                payload = _slice_raw_clip(raw, clip.span) if is_raw else _remux_mp4_clip(raw, clip.span)
+                payload = (_slice_raw_clip(data, clip.span) if is_raw
+                           else _remux_mp4_clip(mp4_bytes, clip.span))
                 arr = np.frombuffer(payload, dtype=np.uint8)
                 clip.encoded_data = LazyData(value=arr, nbytes=arr.nbytes)
                 video.clip_stats.num_transcoded += 1

@@ -157,6 +157,8 @@ class ClipFrameExtractionStage(CuratorStage):
         demuxer's PTS — identical to the raw-NV12 path (§8 row a2).
         """
         lib = hotpath.require_gpu()
+        if not isinstance(data, bytes):
+            data = bytes(data)  # demuxer ABI wants bytes; mp4 payloads are small
         d = hotpath.Demuxer(data)
         try:
             info = d.probe()
@@ -240,7 +242,9 @@ class ClipFrameExtractionStage(CuratorStage):
             if data is None:
                 clip.errors["encoded_data"] = "empty"
                 continue
-            raw = bytes(data) if not isinstance(data, bytes) else data
+            # raw payloads stay numpy (raw_backend consumes any buffer);
+            # the VCN path converts to bytes itself for the demuxer
+            raw = data
             try:
                 local_frames: dict[str, torch.Tensor | np.ndarray] = {}
                 for policy in self._extraction_policies:

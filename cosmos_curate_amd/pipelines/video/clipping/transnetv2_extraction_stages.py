@@ -161,7 +161,7 @@ class VideoFrameExtractionStage(CuratorStage):
                     if data is None:
                         video.errors["frame_extraction"] = "no encoded data"
                         continue
-                    raw = bytes(data) if not isinstance(data, bytes) else data
+                    raw = data  # raw_backend consumes any buffer (no copy)
                     if not raw_backend.is_raw_nv12(raw):
                         video.errors["frame_extraction"] = "decode_unavailable"
                         continue

@@ -93,7 +93,7 @@ class ClipFrameCreationStage(CuratorStage):
                 regen_fps *= 2
                 if regen_fps > 20:
                     break
-                raw = bytes(data) if not isinstance(data, bytes) else data
+                raw = data  # buffers pass through; no bytes() copy
                 frames = extract_frames(
                     raw, sample_rate_fps=regen_fps,
                     target_res=self._target_res,

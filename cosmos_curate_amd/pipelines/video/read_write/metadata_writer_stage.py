@@ -104,9 +104,9 @@ class ClipWriterStage(CuratorStage):
                                else str(clip.uuid))
                         if self._upload_clips and clip.encoded_data:
                             payload = clip.encoded_data.resolve()
-                            (self._output_path / "clips" / f"{cid}.bin").write_bytes(
-                                bytes(payload)
-                            )
+                            with open(self._output_path / "clips" / f"{cid}.bin",
+                                      "wb") as fh:
+                                fh.write(payload)  # buffer protocol: no copy
                         meta = {
                             # "id" is the file/sample id (cam-prefixed for
                             # multicam); downstream consumers (sharding,

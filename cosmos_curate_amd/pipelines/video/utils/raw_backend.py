@@ -23,6 +23,12 @@ import numpy.typing as npt
 
 MAGIC = b"CCNV12RAW\x00"
 _HDR = struct.Struct("<10sHIIIII")
+HEADER_SIZE = _HDR.size
+
+
+def pack_header(n: int, h: int, w: int, fps_num: int, fps_den: int = 1) -> bytes:
+    """The 32-byte payload header alone (for zero-copy body slicing)."""
+    return _HDR.pack(MAGIC, 1, n, h, w, fps_num, fps_den)
 
 
 def encode_raw_nv12(
