@@ -37,23 +37,32 @@ SIGLIP_STD = (0.5, 0.5, 0.5)
 class _CLIPImageEmbeddings(torch.nn.Module):
     """Device-side implementation (reference clip.py:36-74 counterpart)."""
 
-    def __init__(self, variant: str = "vit_b32") -> None:
+    def __init__(self, variant: str = "vit_b32",
+                 checkpoint_path: str | None = None) -> None:
         super().__init__()
+        import os
+
         from cosmos_curate_amd.models import clip_weights as cw
 
         hotpath.require_gpu()  # fail loudly before any lazy surprises
         self.device = torch.device("cuda")
         cfg = cw.CONFIGS[variant]
+        # real checkpoints (reference models/clip.py:33 downloads them;
+        # here the operator provides local files — no network)
+        checkpoint_path = checkpoint_path or os.environ.get("CC_CLIP_CHECKPOINT")
+        pretrained = (cw.load_pretrained_state_dict(checkpoint_path)
+                      if checkpoint_path else None)
         if variant.startswith("siglip"):
             from cosmos_curate_amd.models.siglip_vit import SiglipVisionTowerAMD
 
             self.tower = SiglipVisionTowerAMD(
-                cw.make_siglip_weights(cfg), cfg).to(self.device)
+                pretrained or cw.make_siglip_weights(cfg), cfg).to(self.device)
             self._mean_arr = np.array(SIGLIP_MEAN, dtype=np.float32)
             self._std_arr = np.array(SIGLIP_STD, dtype=np.float32)
         else:
-            sd = (make_clip_vit_b32_weights() if variant == "vit_b32"
-                  else cw.make_clip_vit_weights(cfg))
+            sd = pretrained or (
+                make_clip_vit_b32_weights() if variant == "vit_b32"
+                else cw.make_clip_vit_weights(cfg))
             self.tower = ClipVisionTowerAMD(sd, cfg).to(self.device)
             self._mean_arr = np.array(CLIP_MEAN, dtype=np.float32)
             self._std_arr = np.array(CLIP_STD, dtype=np.float32)
@@ -156,9 +165,11 @@ class CLIPImageEmbeddings(ModelInterface):
     model id, openai/clip-vit-large-patch14 — BASELINE config #3 class).
     """
 
-    def __init__(self, variant: str = "vit_b32") -> None:
+    def __init__(self, variant: str = "vit_b32",
+                 checkpoint_path: str | None = None) -> None:
         super().__init__()
         self._variant = variant
+        self._checkpoint_path = checkpoint_path
         self._model: _CLIPImageEmbeddings | None = None
 
     @property
@@ -174,7 +185,8 @@ class CLIPImageEmbeddings(ModelInterface):
         return [_CLIP_MODEL_ID]
 
     def setup(self) -> None:
-        self._model = _CLIPImageEmbeddings(self._variant)
+        self._model = _CLIPImageEmbeddings(self._variant,
+                                           checkpoint_path=self._checkpoint_path)
 
     def __call__(self, images: torch.Tensor | npt.NDArray[np.uint8]) -> torch.Tensor:
         assert self._model is not None, "setup() not called"
