@@ -765,3 +765,220 @@ extern "C" int cc_gemm_v16(const void* A, const void* B, void* C, long M,
                        ctrs);
   return hipGetLastError() == hipSuccess ? 0 : -1;
 }
+
+// ---- V17: v15 + one-phase-ahead afrag pipeline (no lgkm drains) ----
+// PMC evidence (profiles/r02 pmc_*): 38-42% of wave cycles park at
+// waitcnt/barrier even in the bare kernel.  Each phase issues its
+// ds_reads just before its barrier, so the post-barrier lgkmcnt(0)
+// exposes most of the LDS latency.  v17 double-buffers afrag and issues
+// quadrant q+1's reads BEFORE quadrant q's MFMA cluster (outside the
+// setprio window — v12's mistake was inside it), dropping the explicit
+// lgkm drain entirely: hipcc's register scoreboard emits counted waits
+// for exactly the fragments each MFMA consumes.
+// WAR safety without the lgkm0-before-barrier argument: bfragT reads
+// complete before their ph-q0 MFMA consumes them (compiler wait), 2+
+// phases before STAGE_B2(t+2) overwrites that buffer; A-buf writes
+// target the non-read buffer.  vmcnt(0) publish moved to ph2 (after
+// MFMA(1), before the B(t+2) issues).
+namespace {
+
+template <int PERSIST>
+__global__ __launch_bounds__(512, 1) void k_gemm_v17(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // 128 KiB
+#define A17(b) (lds + (b) * (BM * BK))
+#define B17(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long KT = K / BK;
+  const long srow = wid * 16;
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2][2];  // [pipeline buf][m][kg]
+
+#define STAGE_A17(bmv, t, h)                                                \
+  stage_half15(A, K, (bmv) + (h) * 128 + srow, M, (t) * BK,                 \
+               A17((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B17(bnv, t, h)                                                \
+  stage_half15(B, K, (bnv) + (h) * 128 + srow, N, (t) * BK,                 \
+               B17((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+  const int tiles = PERSIST ? (nwg + (int)gridDim.x - 1) / (int)gridDim.x : 1;
+  long bm = 0, bn = 0;
+  {
+    int orig = (int)blockIdx.x;
+    if (do_remap) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    bm = (long)(orig / nbx) * BM;
+    bn = (long)(orig % nbx) * BN;
+  }
+  STAGE_A17(bm, 0, 0);
+  STAGE_A17(bm, 0, 1);
+  STAGE_B17(bn, 0, 0);
+  STAGE_B17(bn, 0, 1);
+  STAGE_B17(bn, 1, 0);
+  STAGE_B17(bn, 1, 1);
+
+  for (int rep = 0; rep < tiles; rep++) {
+    if (PERSIST && (int)blockIdx.x + rep * (int)gridDim.x >= nwg) break;
+    f32x4 acc[MFR][NFR] = {};
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+#define MFMA17(q, pb)                                                       \
+  do {                                                                      \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[pb][m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0); \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define READ_A17(At, q, pb)                                                 \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[pb][m][g] =         \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B17(Bt)                                                        \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+#define KTILE17(t)                                                          \
+  do {                                                                      \
+    const __bf16* At = A17((t) & 1);                                        \
+    const __bf16* Bt = B17((t) & 1);                                        \
+    READ_B17(Bt);                                                           \
+    READ_A17(At, 0, 0);                                                     \
+    if ((t) + 1 < KT) STAGE_A17(bm, (t) + 1, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    READ_A17(At, 1, 1); /* next quadrant BEFORE this cluster */             \
+    MFMA17(0, 0);                                                           \
+    if ((t) + 1 < KT) STAGE_A17(bm, (t) + 1, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    READ_A17(At, 2, 0);                                                     \
+    MFMA17(1, 1);                                                           \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
+    if ((t) + 2 < KT) STAGE_B17(bn, (t) + 2, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    READ_A17(At, 3, 1);                                                     \
+    MFMA17(2, 0);                                                           \
+    if ((t) + 2 < KT) STAGE_B17(bn, (t) + 2, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    MFMA17(3, 1);                                                           \
+  } while (0)
+
+    for (long it = 0; it < KT / 2; ++it) {
+      KTILE17(2 * it);
+      KTILE17(2 * it + 1);
+    }
+
+    const long ebm = bm, ebn = bn;
+    if (PERSIST && rep + 1 < tiles &&
+        (int)blockIdx.x + (rep + 1) * (int)gridDim.x < nwg) {
+      __builtin_amdgcn_s_barrier();
+      int orig = (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (do_remap) {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = orig & 7, lid = orig >> 3;
+        orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+      }
+      bm = (long)(orig / nbx) * BM;
+      bn = (long)(orig % nbx) * BN;
+      STAGE_A17(bm, 0, 0);
+      STAGE_A17(bm, 0, 1);
+      STAGE_B17(bn, 0, 0);
+      STAGE_B17(bn, 0, 1);
+      STAGE_B17(bn, 1, 0);
+      STAGE_B17(bn, 1, 1);
+    }
+
+    const long crow_base = ebm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN + (lane & 15);
+    const bool interior = (ebm + BM <= M) && (ebn + BN <= N);
+    if (interior) {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (col >= N) continue;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (row >= M) continue;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    }
+  }
+#undef A17
+#undef B17
+#undef STAGE_A17
+#undef STAGE_B17
+#undef MFMA17
+#undef READ_A17
+#undef READ_B17
+#undef KTILE17
+}
+
+}  // namespace
+
+extern "C" int cc_gemm_v17(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int persist,
+                           int remap, unsigned long long stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4) return -2;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg;
+  if (persist) grid = nwg < 256 ? nwg : 256;
+  if (persist)
+    hipLaunchKernelGGL((k_gemm_v17<1>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  else
+    hipLaunchKernelGGL((k_gemm_v17<0>), dim3(grid), dim3(512), 0,
+                       (hipStream_t)stream, (const __bf16*)A,
+                       (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,
+                       remap);
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
