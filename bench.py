@@ -113,7 +113,13 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--clips", type=int, default=64, help="clips per step per rank")
+    ap.add_argument(
+        "--clips", type=int, default=160,
+        help="clips per step per rank (default 160: the persistent GEMM "
+        "amortizes per-tile costs over ~2-3x more tiles per WG than the "
+        "round-1 batch of 64; measured plateau 3980+ clips/s at 160-256, "
+        "profiles/r02_batch_sweep*.log)",
+    )
     ap.add_argument(
         "--host-fed", action="store_true",
         help="measurement leg: NV12 starts in PINNED HOST memory and the "
