@@ -174,10 +174,8 @@ class ClipFrameExtractionStage(CuratorStage):
                 pkt, tick, _ = d.packet(s)
                 pkts.append(pkt)
                 pts_ticks[s] = tick
-            order = np.argsort(pts_ticks, kind="stable")  # presentation order
             # sampled presentation indices -> wanted pts ticks -> slots
-            slot_by_tick = {int(pts_ticks[order[i]]): j
-                            for j, i in enumerate(idx)}
+            slot_by_tick = vcn_slot_map(pts_ticks, idx)
 
             dev = torch.device("cuda")
             stream = torch.cuda.current_stream(dev).cuda_stream
@@ -295,6 +293,19 @@ class ClipFrameExtractionStage(CuratorStage):
                 name, stats = self._timer.log_stats()
                 task.stage_perf[name] = stats
         return tasks
+
+def vcn_slot_map(pts_ticks: "np.ndarray", sampled_idx: "np.ndarray") -> dict[int, int]:
+    """Map pts tick -> output slot for sampled PRESENTATION indices.
+
+    ``pts_ticks`` are per-sample pts in DECODE order (cc_demux_packet
+    walk); ``sampled_idx`` indexes the SORTED presentation timeline
+    (sample_closest's contract over get_video_timestamps).  The VCN
+    session emits frames in display order tagged with their pts, so the
+    lookup key is the tick value itself.  Pure function, unit-tested
+    against B-frame-style reorderings (test_stages_cpu)."""
+    order = np.argsort(pts_ticks, kind="stable")  # presentation order
+    return {int(pts_ticks[order[i]]): j for j, i in enumerate(sampled_idx)}
+
 
 def extract_frames(
     data: bytes,

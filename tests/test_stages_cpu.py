@@ -288,3 +288,35 @@ def test_frame_creation_reextraction_passes_target_res(monkeypatch):
     stage.process_data([SplitPipeTask(videos=[Video(input_video="v", clips=[clip])])])
     assert seen_kwargs.get("target_res") == (64, 64)
     assert len(clip.clip_embedding_frames.resolve()) == 8
+
+
+def test_vcn_slot_map_bframe_reordering():
+    """The VCN path's pts->slot mapping on a B-frame-style decode order
+    (decode order != presentation order), against sample_closest's
+    sorted-timeline contract."""
+    import numpy as np
+
+    from cosmos_curate_amd.pipelines.video.clipping.clip_frame_extraction_stages import (
+        vcn_slot_map,
+    )
+    from cosmos_curate_amd.pipelines.video.utils.decoder_utils import sample_closest
+
+    # decode order: I P B B P B B ... (pts ticks at 512/frame, shifted)
+    pres_ticks = 1024 + 512 * np.arange(12, dtype=np.int64)
+    # pts_ticks[k] = presentation tick of the k-th frame in DECODE order
+    decode_order = np.array([0, 3, 1, 2, 6, 4, 5, 9, 7, 8, 11, 10])
+    pts_ticks = pres_ticks[decode_order]
+    # presentation-sorted seconds timeline (timescale 12288)
+    ts = np.sort((pts_ticks / 12288.0).astype(np.float32))
+    idx, counts, _ = sample_closest(ts, sample_rate=12288.0 / 512 / 2)
+
+    m = vcn_slot_map(pts_ticks, idx)
+    # the mapping must point each WANTED presentation position's tick at
+    # its output slot, regardless of decode order
+    assert len(m) == len(idx)
+    for j, i in enumerate(idx):
+        tick = int(round(float(ts[i]) * 12288))
+        assert m[tick] == j
+    # and ticks NOT sampled are absent (frames get skipped, not written)
+    all_ticks = set(int(t) for t in pts_ticks)
+    assert set(m).issubset(all_ticks)
