@@ -15,6 +15,8 @@ shapes = {
     "fc1": [(67200, 3072, 768, 1, 0)],     # act=1 (quick-gelu), no res
     "out": [(67200, 768, 768, 0, 1)],      # residual fused
     "bare_fc1": [(67200, 3072, 768, 0, -1)],  # no bias, no res (screen cfg)
+    "fc1_b160": [(168000, 3072, 768, 1, 0)],  # bench default batch 160
+    "out_b160": [(168000, 768, 768, 0, 1)],
 }
 for (M, N, K, act, res) in shapes[os.environ.get("CC_PMC_SHAPE", "fc1")]:
     torch.manual_seed(1)
