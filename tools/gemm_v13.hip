@@ -982,3 +982,225 @@ extern "C" int cc_gemm_v17(const void* A, const void* B, void* C, long M,
                        remap);
   return hipGetLastError() == hipSuccess ? 0 : -1;
 }
+
+// ---- V18: ablation skeleton of the v15 persistent loop ----
+// Wrong-results-by-design cost decomposition (guide §5 mistake 8:
+// ablate empirically).  ABLATE: 0 = full kernel; 1 = no glds staging
+// and no vmcnt drains (stale LDS); 2 = no fragment ds_reads (stale
+// regs); 3 = both; 4 = no barriers (races, full compute); 5 = MFMA +
+// barriers only.  Only mode 0 computes C correctly.
+namespace {
+
+template <int ABLATE>
+__global__ __launch_bounds__(512, 1) void k_gemm_v18(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];
+#define A18(b) (lds + (b) * (BM * BK))
+#define B18(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  constexpr bool DO_STAGE = ABLATE == 0 || ABLATE == 2 || ABLATE == 4;
+  constexpr bool DO_READS = ABLATE == 0 || ABLATE == 1 || ABLATE == 4;
+  constexpr bool DO_BARRIER = ABLATE != 4 && ABLATE != 7;
+  constexpr bool DO_PRIO = ABLATE != 6 && ABLATE != 7 && ABLATE != 8 && ABLATE != 9;
+  // modes 8/9: thin the barriers to 2 / 1 per K-tile (skeleton only)
+  constexpr int BAR_EVERY = ABLATE == 8 ? 2 : (ABLATE == 9 ? 4 : 1);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long KT = K / BK;
+  const long srow = wid * 16;
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2] = {};
+  bf16x8 afrag[2][2] = {};
+
+#define STAGE_A18(t, h)                                                     \
+  if constexpr (DO_STAGE)                                                   \
+  stage_half15(A, K, bm + (h) * 128 + srow, M, (t) * BK,                    \
+               A18((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B18(t, h)                                                     \
+  if constexpr (DO_STAGE)                                                   \
+  stage_half15(B, K, bn + (h) * 128 + srow, N, (t) * BK,                    \
+               B18((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define BAR18()                                                             \
+  if constexpr (DO_BARRIER) __builtin_amdgcn_s_barrier()
+#define BARN18(ph)                                                          \
+  if constexpr (DO_BARRIER) {                                               \
+    if constexpr ((ph) % BAR_EVERY == 0) __builtin_amdgcn_s_barrier();      \
+  }
+#define DRAIN18()                                                           \
+  if constexpr (DO_STAGE) asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
+
+  const int tiles = (nwg + (int)gridDim.x - 1) / (int)gridDim.x;
+  long bm = 0, bn = 0;
+  {
+    int orig = (int)blockIdx.x;
+    if (do_remap) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    bm = (long)(orig / nbx) * BM;
+    bn = (long)(orig % nbx) * BN;
+  }
+  STAGE_A18(0, 0);
+  STAGE_A18(0, 1);
+  STAGE_B18(0, 0);
+  STAGE_B18(0, 1);
+  STAGE_B18(1, 0);
+  STAGE_B18(1, 1);
+
+  for (int rep = 0; rep < tiles; rep++) {
+    if ((int)blockIdx.x + rep * (int)gridDim.x >= nwg) break;
+    f32x4 acc[MFR][NFR] = {};
+    DRAIN18();
+    BAR18();
+
+#define PH18(q)                                                             \
+  do {                                                                      \
+    if constexpr (DO_READS)                                                 \
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                    \
+    if constexpr (DO_PRIO) __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+    if constexpr (DO_PRIO) __builtin_amdgcn_s_setprio(0);                   \
+  } while (0)
+
+#define RA18(At, q)                                                         \
+  if constexpr (DO_READS) {                                                 \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      const int k16 = (g << 2) + (lane >> 4);                               \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =           \
+          frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);              \
+    }                                                                       \
+  }
+
+#define RB18(Bt)                                                            \
+  if constexpr (DO_READS) {                                                 \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      const int k16 = (g << 2) + (lane >> 4);                               \
+      _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =        \
+          frag_read(Bt, brow_base + n * FRAG, k16);                         \
+    }                                                                       \
+  }
+
+#define KT18(t)                                                             \
+  do {                                                                      \
+    const __bf16* At = A18((t) & 1);                                        \
+    const __bf16* Bt = B18((t) & 1);                                        \
+    RB18(Bt);                                                               \
+    RA18(At, 0);                                                            \
+    if ((t) + 1 < KT) STAGE_A18((t) + 1, 0);                                \
+    BARN18(0);                                                              \
+    PH18(0);                                                                \
+    RA18(At, 1);                                                            \
+    if ((t) + 1 < KT) STAGE_A18((t) + 1, 1);                                \
+    BARN18(1);                                                              \
+    PH18(1);                                                                \
+    RA18(At, 2);                                                            \
+    DRAIN18();                                                              \
+    if ((t) + 2 < KT) STAGE_B18((t) + 2, 0);                                \
+    BARN18(2);                                                              \
+    PH18(2);                                                                \
+    RA18(At, 3);                                                            \
+    if ((t) + 2 < KT) STAGE_B18((t) + 2, 1);                                \
+    BARN18(3);                                                              \
+    PH18(3);                                                                \
+  } while (0)
+
+    for (long it = 0; it < KT / 2; ++it) {
+      KT18(2 * it);
+      KT18(2 * it + 1);
+    }
+
+    const long ebm = bm, ebn = bn;
+    if (rep + 1 < tiles &&
+        (int)blockIdx.x + (rep + 1) * (int)gridDim.x < nwg) {
+      BAR18();
+      int orig = (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (do_remap) {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = orig & 7, lid = orig >> 3;
+        orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+      }
+      bm = (long)(orig / nbx) * BM;
+      bn = (long)(orig % nbx) * BN;
+      STAGE_A18(0, 0);
+      STAGE_A18(0, 1);
+      STAGE_B18(0, 0);
+      STAGE_B18(0, 1);
+      STAGE_B18(1, 0);
+      STAGE_B18(1, 1);
+    }
+
+    const long crow_base = ebm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN + (lane & 15);
+#pragma unroll
+    for (int m = 0; m < MFR; m++) {
+#pragma unroll
+      for (int n = 0; n < NFR; n++) {
+        const long col = ccol_base + n * FRAG;
+        if (col >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const long row = crow_base + m * FRAG + r;
+          if (row >= M) continue;
+          if (c_is_bf16)
+            ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+          else
+            ((float*)C)[row * N + col] = acc[m][n][r];
+        }
+      }
+    }
+  }
+#undef A18
+#undef B18
+#undef STAGE_A18
+#undef STAGE_B18
+#undef BAR18
+#undef DRAIN18
+#undef PH18
+#undef RA18
+#undef RB18
+#undef KT18
+}
+
+}  // namespace
+
+extern "C" int cc_gemm_v18(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int ablate,
+                           int remap, unsigned long long stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4) return -2;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg < 256 ? nwg : 256;
+#define L18(AB)                                                            \
+  hipLaunchKernelGGL((k_gemm_v18<AB>), dim3(grid), dim3(512), 0,           \
+                     (hipStream_t)stream, (const __bf16*)A,                \
+                     (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,    \
+                     remap)
+  switch (ablate) {
+    case 0: L18(0); break;
+    case 1: L18(1); break;
+    case 2: L18(2); break;
+    case 3: L18(3); break;
+    case 4: L18(4); break;
+    case 5: L18(5); break;
+    case 6: L18(6); break;
+    case 7: L18(7); break;
+    case 8: L18(8); break;
+    case 9: L18(9); break;
+    default: return -3;
+  }
+#undef L18
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
