@@ -552,19 +552,28 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(1);                                                         \
     READ_A2(At, 2);                                                         \
-    /* full drain BEFORE issuing B(t+2): everything ph5 needs (A(t+1),   */ \
-    /* B(t+1)) has been issued and must land; a COUNTED wait is unsound  */ \
-    /* here because VMEM completions can retire out of order under load  */ \
-    /* (measured: vmcnt(4) at ph4 was nondeterministic at >=774-block    */ \
-    /* grids, profiles/r01_t256_det2.log); the drain sits one phase      */ \
-    /* before the consumer and ahead of the B(t+2) issues, so 2 half-    */ \
-    /* tiles still span the tile boundary.                               */ \
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                        \
-    if ((t) + 2 < KT) STAGE_B2((t) + 2, 0);                                 \
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(2);                                                         \
     READ_A2(At, 3);                                                         \
-    if ((t) + 2 < KT) STAGE_B2((t) + 2, 1);                                 \
+    /* COUNTED publish (round 2, replaces the per-tile vmcnt(0) drain):  */ \
+    /* the next tile's LDS reads need A(t+1) [ph1-2, 4 loads/wave] and   */ \
+    /* B(t+1) [issued one tile earlier, strictly older] landed.  After   */ \
+    /* issuing B(t+2)'s 4 loads, the per-wave VMEM queue newest-first is */ \
+    /* [B(t+2):4, A(t+1):4, <older>]; vmcnt(4) retires everything except */ \
+    /* B(t+2), which keeps spanning the tile boundary.  r01's            */ \
+    /* "nondeterministic counted wait" (profiles/r01_t256_det2.log) sat  */ \
+    /* BEFORE the B(t+2) issues with exactly 4 loads outstanding — a     */ \
+    /* no-op wait, not out-of-order retirement; this placement fixes the */ \
+    /* count.  Verified: 6-run multi-shape determinism screen + full GPU */ \
+    /* parity suite (profiles/r02_gemm_v19.log).  Measured: fc1 864->905,*/ \
+    /* fc2 950->1023, patch 940->1004 TF within one box.                 */ \
+    if ((t) + 2 < KT) {                                                     \
+      STAGE_B2((t) + 2, 0);                                                 \
+      STAGE_B2((t) + 2, 1);                                                 \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                      \
+    } else {                                                                \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
+    }                                                                       \
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(3);                                                         \
   } while (0)
