@@ -112,7 +112,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int do_remap) {
+    int c_is_bf16, int nbx, int nwg, int do_remap, int epi_staged) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];  // one __shared__ object (G16 4a)
 #define AS(b) (lds + (b) * (BM * BK))
 #define BS(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -270,7 +270,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int do_remap) {
+    int c_is_bf16, int nbx, int nwg, int do_remap, int epi_staged) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];
 #define AS32(b) (lds + (b) * (BM * BK))
 #define BS32(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -440,7 +440,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int order_mode) {
+    int c_is_bf16, int nbx, int nwg, int order_mode, int epi_staged) {
   // Round-2 structure: PERSISTENT multi-tile (grid = min(nwg, 256), one
   // WG per CU) around the round-1 8-phase glds schedule.  Each WG walks
   // tiles bid, bid+grid, ... and issues the NEXT tile's 12-glds
@@ -626,7 +626,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     // the scalar path, inside the embedding-cosine contract.
     const long crow_base = ebm + waveM * WM2 + 4 * (lane >> 4);
     const long ccol_base = ebn + waveN * WN2 + (lane & 15);
-    if (c_is_bf16) {
+    if (c_is_bf16 && epi_staged) {
       __bf16* slice = A2T(1) + wid * 2048;  // 4 KB per wave
       const long wrow0 = ebm + waveM * WM2;
       const long wcol0 = ebn + waveN * WN2;
@@ -671,7 +671,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
         }
       }
     } else {
-      // f32 output (visual projection / probes): scalar store path
+      // scalar store path (f32 outputs always; bf16 when epi_staged=0)
       const bool interior = (ebm + BM2 <= M) && (ebn + BN2 <= N);
 #pragma unroll
       for (int m = 0; m < MFR2; m++) {
@@ -687,7 +687,10 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
             if (!interior && row >= M) continue;
             float v = cc_act<ACT>(acc[m][n][r] + bval);
             if constexpr (HAS_RES) v += (float)residual[row * N + col];
-            ((float*)C)[row * N + col] = v;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = f32_to_bf16_rne(v);
+            else
+              ((float*)C)[row * N + col] = v;
           }
         }
       }
@@ -775,11 +778,22 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
     // wins (square8k 1295 vs 944 — profiles/r02_gemm_persist.log)
     remap = nbx <= 16 ? 1 : 0;
   }
+  // bf16 epilogue: LDS-staged coalesced when the launch fuses bias/act/
+  // residual (A/B: tower 582 -> 812 TF, bench 3184 -> 4188 clips/s,
+  // profiles/r02_epi_ab.log), scalar for bare C=A*B^T launches where the
+  // 4-pass staging is pure overhead (bare fc1 928 scalar vs 825 staged,
+  // profiles/r02_gemm_v20.log).  CC_GEMM_EPI=0|1 forces it for A/B.
+  static const int epi_env = [] {
+    const char* e = getenv("CC_GEMM_EPI");
+    return e ? atoi(e) : -1;
+  }();
+  const int epi_auto = (bias != nullptr || residual != nullptr || act != 0) ? 1 : 0;
+  const int epi_staged = epi_env >= 0 ? epi_env : epi_auto;
 #define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
   hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \
                      (const __bf16*)A, (const __bf16*)B, C, bias,             \
                      (const __bf16*)residual, (long)M, (long)N, (long)K,      \
-                     c_dtype == 1 ? 1 : 0, nbx, nwg, remap)
+                     c_dtype == 1 ? 1 : 0, nbx, nwg, remap, epi_staged)
 #define CC_DISPATCH_ACT(KER, A_)                                              \
   do {                                                                        \
     if (hb && hr) CC_LAUNCH_GEMM(KER, A_, true, true);                        \
