@@ -542,6 +542,13 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
   do {                                                                      \
     const __bf16* At = A2T((t) & 1);                                        \
     const __bf16* Bt = B2T((t) & 1);                                        \
+    /* counted publish for THIS tile's operands, placed as late as      */ \
+    /* possible (after the previous tile's MFMA(3) instead of before    */ \
+    /* it): newest-first per-wave queue here is [B(t+1):4, A(t):4, ...] */ \
+    /* so vmcnt(4) retires A(t)/B(t) and keeps B(t+1) in flight.  At    */ \
+    /* tile 0 of a rep the queue is already drained (rep-top vmcnt(0))  */ \
+    /* and this is a no-op.                                             */ \
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                        \
     READ_B2(Bt);                                                            \
     READ_A2(At, 0);                                                         \
     if ((t) + 1 < KT) STAGE_A2((t) + 1, 0);                                 \
@@ -570,9 +577,10 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     if ((t) + 2 < KT) {                                                     \
       STAGE_B2((t) + 2, 0);                                                 \
       STAGE_B2((t) + 2, 1);                                                 \
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                      \
+      /* publish moved to the NEXT tile's top (one more MFMA cluster of  */ \
+      /* latency cover at the same count)                                */ \
     } else {                                                                \
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); /* tail drain */     \
     }                                                                       \
     __builtin_amdgcn_s_barrier();                                           \
     PHASE_MFMA2(3);                                                         \
@@ -630,6 +638,16 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
       __bf16* slice = A2T(1) + wid * 2048;  // 4 KB per wave
       const long wrow0 = ebm + waveM * WM2;
       const long wcol0 = ebn + waveN * WN2;
+      // bias depends only on the n-fragment: 4 loads per lane, hoisted
+      // out of the 4x2 pass loop (was reloaded 8x)
+      float bhoist[NFR2] = {};
+      if constexpr (HAS_BIAS) {
+#pragma unroll
+        for (int n = 0; n < NFR2; n++) {
+          const long col = ccol_base + n * FRAG;
+          bhoist[n] = (col < N) ? bias[col] : 0.0f;
+        }
+      }
 #pragma unroll
       for (int p = 0; p < 4; p++) {
 #pragma unroll
@@ -637,9 +655,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
           const int m = 2 * p + mm;
 #pragma unroll
           for (int n = 0; n < NFR2; n++) {
-            const long col = ccol_base + n * FRAG;
-            float bval = 0.0f;
-            if constexpr (HAS_BIAS) bval = (col < N) ? bias[col] : 0.0f;
+            const float bval = bhoist[n];
             const int lrow_b = mm * 16 + 4 * (lane >> 4);
             const int lcol = n * FRAG + (lane & 15);
 #pragma unroll
