@@ -114,11 +114,11 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument(
-        "--clips", type=int, default=160,
-        help="clips per step per rank (default 160: the persistent GEMM "
-        "amortizes per-tile costs over ~2-3x more tiles per WG than the "
-        "round-1 batch of 64; measured plateau 3980+ clips/s at 160-256, "
-        "profiles/r02_batch_sweep*.log)",
+        "--clips", type=int, default=224,
+        help="clips per step per rank (default 224: the persistent GEMM "
+        "amortizes per-tile costs over more tiles per WG; clips/s "
+        "plateaus from batch 160 and the GEMM roofline fraction peaks "
+        "around 224 — profiles/r02_batch_sweep*.log, r02_batch3.log)",
     )
     ap.add_argument(
         "--host-fed", action="store_true",
