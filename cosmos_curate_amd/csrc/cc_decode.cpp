@@ -301,8 +301,12 @@ int cc_decode_map_frames(cc_decode* s, cc_nv12_frame* out, size_t cap,
                            disp.picture_index, rocdec_err(s->api, st));
     s->display_q.pop_front();
     cc_nv12_frame* f = &out[(*n)++];
-    f->y = planes[0];
-    f->uv = planes[1];
+    // apply the display crop origin (usually 0,0; e.g. 1920x1088 coded
+    // with display_area {0,0,1920,1080}): callers see display pixels.
+    const int left = s->format.display_area.left & ~1;  // NV12: even x
+    const int top = s->format.display_area.top & ~1;
+    f->y = (uint8_t*)planes[0] + (size_t)top * pitch[0] + left;
+    f->uv = (uint8_t*)planes[1] + (size_t)(top / 2) * pitch[1] + left;
     f->pitch = pitch[0];
     f->pts = (int64_t)disp.pts;
     f->width = (uint32_t)(s->format.display_area.right -
