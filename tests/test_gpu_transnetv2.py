@@ -1,6 +1,7 @@
 """TransNetV2 split path end-to-end on MI355X (raw-NV12 backend)."""
 
 import pathlib
+import time
 
 import numpy as np
 import pytest
@@ -81,3 +82,70 @@ def test_transnetv2_pipeline_end_to_end():
     gpu_out = m_gpu(x.cuda())[0, :, 0].cpu().numpy()
     np.testing.assert_allclose(gpu_out, cpu_out, atol=5e-4)
     np.testing.assert_allclose(gpu_out, g["output"], atol=5e-4)
+
+
+def test_4k_stream_shot_boundary_to_embeddings():
+    """Config #4's per-stream core at 4K coded size (3840x2160): shot
+    boundary (TransNetV2) split -> clip frame extraction (4K -> 224) ->
+    CLIP embeddings, end to end on device, with bit-exact 27x48 frame
+    parity spot-checked against the pixel oracle at 4K dims."""
+    from cosmos_curate_amd.pipelines.video.clipping.clip_extraction_stages import (
+        ClipTranscodingStage,
+    )
+    from cosmos_curate_amd.pipelines.video.clipping.clip_frame_extraction_stages import (
+        ClipFrameExtractionStage,
+    )
+    from cosmos_curate_amd.pipelines.video.clipping.transnetv2_extraction_stages import (
+        TransNetV2ClipExtractionStage,
+    )
+    from cosmos_curate_amd.pipelines.video.embedding.clip_stages import (
+        ClipEmbeddingStage,
+        ClipFrameCreationStage,
+    )
+
+    h4k, w4k, fps, secs = 2160, 3840, 30, 2
+    raw = raw_backend.make_synthetic_clip(fps * secs, h4k, w4k, fps, seed=7)
+    v = Video(
+        input_video=pathlib.Path("/synthetic/4k.mp4"),
+        metadata=VideoMetadata(size=1, height=h4k, width=w4k,
+                               framerate=float(fps), num_frames=fps * secs,
+                               duration=float(secs), video_codec="raw"),
+        encoded_data=np.frombuffer(raw, dtype=np.uint8),
+    )
+    task = SplitPipeTask(videos=[v])
+
+    t0 = time.perf_counter()
+    out = run_pipeline([task], [VideoFrameExtractionStage()],
+                       runner=SequentialRunner())
+    # 27x48 parity at 4K dims (the config-#4 shot-boundary input path);
+    # checked here because the transnet stage drops frame_array after use
+    frames = out[0].video.frame_array.resolve()
+    assert frames.shape == (fps * secs, 27, 48, 3)
+    ys, uvs = raw_backend.frame_planes(raw, np.array([0, 30], dtype=np.int32))
+    for j, i in enumerate([0, 30]):
+        rgb = ocolor.nv12_to_rgb(ys[j], uvs[j].reshape(h4k // 2, w4k // 2, 2))
+        want = ocolor.resize_bilinear_u8(rgb, 27, 48)
+        np.testing.assert_array_equal(frames[i], want)
+
+    out = run_pipeline(
+        out,
+        [TransNetV2ClipExtractionStage(min_length_s=0.5, min_length_frames=8,
+                                       crop_s=None),
+         ClipTranscodingStage(),
+         ClipFrameExtractionStage(target_res=(224, 224)),
+         ClipFrameCreationStage(target_fps=2.0),
+         ClipEmbeddingStage()],
+        runner=SequentialRunner(),
+    )
+    dt = time.perf_counter() - t0
+    video = out[0].video
+
+    # shot-boundary clips exist and every clip carries an embedding
+    assert video.clips, "no clips from shot-boundary split"
+    for clip in video.clips:
+        assert not clip.errors, clip.errors
+        assert clip.clip_embedding is not None
+        assert clip.clip_embedding.shape == (512,)
+        n = float(np.linalg.norm(clip.clip_embedding))
+        assert abs(n - 1.0) < 1e-3
+    print(f"4k e2e: {len(video.clips)} clips from {secs}s in {dt:.2f}s")
