@@ -687,16 +687,26 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
         }
       }
     } else {
-      // scalar store path (f32 outputs always; bf16 when epi_staged=0)
+      // scalar store path (f32 outputs always; bf16 when epi_staged=0).
+      // bias hoisted exactly like the staged path: 4 loads per lane
+      // instead of one dependent load per fragment (the old
+      // per-element-load serialization, guide trap (c)).
       const bool interior = (ebm + BM2 <= M) && (ebn + BN2 <= N);
+      float bhoist2[NFR2] = {};
+      if constexpr (HAS_BIAS) {
+#pragma unroll
+        for (int n = 0; n < NFR2; n++) {
+          const long col = ccol_base + n * FRAG;
+          bhoist2[n] = (interior || col < N) ? bias[col] : 0.0f;
+        }
+      }
 #pragma unroll
       for (int m = 0; m < MFR2; m++) {
 #pragma unroll
         for (int n = 0; n < NFR2; n++) {
           const long col = ccol_base + n * FRAG;
           if (!interior && col >= N) continue;
-          float bval = 0.0f;
-          if constexpr (HAS_BIAS) bval = bias[col];
+          const float bval = bhoist2[n];
 #pragma unroll
           for (int r = 0; r < 4; r++) {
             const long row = crow_base + m * FRAG + r;
