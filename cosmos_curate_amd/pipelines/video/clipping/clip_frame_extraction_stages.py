@@ -198,6 +198,13 @@ class ClipFrameExtractionStage(CuratorStage):
                         if rgb is None:
                             rgb = torch.empty((n_sel, oh, ow, 3),
                                               dtype=torch.uint8, device=dev)
+                        elif rgb.shape[1:3] != (oh, ow):
+                            # mid-stream resolution change with no target
+                            # resolution: no consistent output shape exists
+                            msg = (f"mid-clip resolution change "
+                                   f"{tuple(rgb.shape[1:3])} -> {(oh, ow)} "
+                                   "with target_res unset")
+                            raise RuntimeError(msg)
                         hotpath.check(lib.cc_nv12_to_rgb_resize(
                             f.y, f.uv, 1, fh, fw, f.pitch,
                             rgb[slot].data_ptr(), oh, ow, stream,
