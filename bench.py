@@ -45,7 +45,10 @@ def gemm_flops_per_frame(variant: str = "vit_b32") -> float:
     cfg = cw.CONFIGS[variant]
     g = cfg.image // cfg.patch
     patches, tokens = g * g, cfg.num_pos  # num_pos handles no-CLS towers
-    patch_k = (3 * cfg.patch * cfg.patch + 63) // 64 * 64  # kernel K padding
+    # ALGORITHMIC K for the patch embed (3*patch^2), NOT the kernel's
+    # 64-multiple padding: padded columns multiply zeros and must not
+    # inflate `achieved` (tier contract: algorithmic flops only).
+    patch_k = 3 * cfg.patch * cfg.patch
     per_layer = (
         2 * tokens * cfg.hidden * (3 * cfg.hidden)  # fused qkv
         + 2 * tokens * cfg.hidden * cfg.hidden      # out proj
