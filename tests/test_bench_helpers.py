@@ -15,7 +15,7 @@ def test_gemm_flops_accounting_matches_hand_calc():
     l14 = bench.gemm_flops_per_frame("vit_l14")
     per_layer14 = 2 * 257 * 1024 * 3072 + 2 * 257 * 1024 * 1024 \
         + 2 * 257 * 1024 * 4096 + 2 * 257 * 4096 * 1024
-    assert l14 == float(2 * 256 * 640 * 1024 + 24 * per_layer14 + 2 * 1024 * 768)
+    assert l14 == float(2 * 256 * 588 * 1024 + 24 * per_layer14 + 2 * 1024 * 768)
 
 
 def test_make_nv12_batch_shape_and_determinism():
