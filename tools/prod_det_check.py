@@ -12,14 +12,15 @@ def main():
     lib = hotpath.require_gpu()
     stream = torch.cuda.current_stream().cuda_stream
     shapes = [(65856, 768, 3072), (67200, 2304, 768), (67200, 3072, 768),
-              (67200, 768, 768), (4096, 4096, 4096)]
+              (67200, 768, 768), (4096, 4096, 4096), (235200, 3072, 768),
+              (230496, 768, 3072)]
     bad = 0
     for (M, N, K) in shapes:
         torch.manual_seed(7)
         a = (torch.randn(M, K) * 0.3).to(torch.bfloat16).cuda()
         b = (torch.randn(N, K) * 0.3).to(torch.bfloat16).cuda()
         ref = None
-        for run in range(6):
+        for run in range(10):
             c = torch.full((M, N), 9.0, dtype=torch.bfloat16, device="cuda")
             hotpath.check(lib.cc_gemm_bf16(a.data_ptr(), b.data_ptr(),
                                            c.data_ptr(), M, N, K, None, 1,
