@@ -503,7 +503,17 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
 #define PHASE_MFMA2(q)                                                      \
   do {                                                                      \
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
-    __builtin_amdgcn_s_setprio(1);                                          \
+    /* T19 dictation (round 2, +1-3%/shape, profiles/r02_gemm_v21.log):  */ \
+    /* weave the next phase's ds_reads and the glds issues between the   */ \
+    /* 16 MFMAs so their latency hides under the cluster (the source-    */ \
+    /* level version of this, v12, regressed; the compile-time directive */ \
+    /* does it without perturbing issue order elsewhere).  setprio       */ \
+    /* dropped: it fenced the scheduler out of exactly this interleave.  */ \
+    _Pragma("unroll") for (int gi = 0; gi < 4; gi++) {                      \
+      __builtin_amdgcn_sched_group_barrier(0x8 /*MFMA*/, 4, 0);             \
+      __builtin_amdgcn_sched_group_barrier(0x100 /*DS_READ*/, 1, 0);        \
+      __builtin_amdgcn_sched_group_barrier(0x10 /*VMEM*/, 1, 0);            \
+    }                                                                       \
     _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
       _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
         _Pragma("unroll") for (int n = 0; n < NFR2; n++) {                  \
@@ -512,7 +522,6 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
         }                                                                   \
       }                                                                     \
     }                                                                       \
-    __builtin_amdgcn_s_setprio(0);                                          \
   } while (0)
   /* NOTE: no trailing barrier per phase (v11, profiles/r01_v11_ab.log,
      +5-8%): the B-buffer write-after-read hazard needs only 2-phase

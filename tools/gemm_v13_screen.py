@@ -65,6 +65,8 @@ def build() -> ctypes.CDLL:
     lib.cc_gemm_v19.restype = ctypes.c_int
     lib.cc_gemm_v20.argtypes = lib.cc_gemm_v13.argtypes
     lib.cc_gemm_v20.restype = ctypes.c_int
+    lib.cc_gemm_v21.argtypes = lib.cc_gemm_v13.argtypes
+    lib.cc_gemm_v21.restype = ctypes.c_int
     return lib
 
 
@@ -86,7 +88,8 @@ def refcheck(prod, v13, stream) -> int:
             for name, fn in (("v13", v13.cc_gemm_v13), ("v15", v13.cc_gemm_v15),
                              ("v17", v13.cc_gemm_v17),
                              ("v19", v13.cc_gemm_v19),
-                             ("v20", v13.cc_gemm_v20), ("v16", None)):
+                             ("v20", v13.cc_gemm_v20),
+                             ("v21", v13.cc_gemm_v21), ("v16", None)):
                 c2.fill_(7.0)
                 if name == "v16":
                     rc = v13.cc_gemm_v16(a.data_ptr(), b.data_ptr(),
@@ -121,6 +124,7 @@ def determinism(v13, stream) -> int:
                                  ("v17", v13.cc_gemm_v17),
                                  ("v19", v13.cc_gemm_v19),
                                  ("v20", v13.cc_gemm_v20),
+                                 ("v21", v13.cc_gemm_v21),
                                  ("v16", None)):
                     if name == "v16" and remap > 1:
                         continue
@@ -154,7 +158,7 @@ def determinism(v13, stream) -> int:
 def perf(prod, v13, stream) -> None:
     ev0, ev1 = torch.cuda.Event(True), torch.cuda.Event(True)
     print(f"{'shape':9s} {'M':>6s} {'N':>5s} {'K':>5s} "
-          f"{'prod':>7s} {'v13':>7s} {'v13p':>7s} {'v13pr':>7s} {'v15p':>7s} {'v15pr':>7s} {'v15pc':>7s} {'v16g':>7s} {'v16x':>7s} {'v17p':>7s} {'v17pr':>7s} {'v19p':>7s} {'v19pr':>7s} {'v20p':>7s} {'v20pr':>7s}")
+          f"{'prod':>7s} {'v13':>7s} {'v13p':>7s} {'v13pr':>7s} {'v15p':>7s} {'v15pr':>7s} {'v15pc':>7s} {'v16g':>7s} {'v16x':>7s} {'v17p':>7s} {'v17pr':>7s} {'v19p':>7s} {'v19pr':>7s} {'v20p':>7s} {'v20pr':>7s} {'v21p':>7s} {'v21pr':>7s}")
     for label, M, N, K in B64:
         torch.manual_seed(1)
         a = torch.randn(M, K).to(torch.bfloat16).cuda()
@@ -219,11 +223,17 @@ def perf(prod, v13, stream) -> None:
         tf_20pr = time_one(lambda: v13.cc_gemm_v20(
             a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, 1, 1,
             stream))
+        tf_21p = time_one(lambda: v13.cc_gemm_v21(
+            a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, 1, 0,
+            stream))
+        tf_21pr = time_one(lambda: v13.cc_gemm_v21(
+            a.data_ptr(), b.data_ptr(), c.data_ptr(), M, N, K, 1, 1, 1,
+            stream))
         print(f"{label:9s} {M:6d} {N:5d} {K:5d} {tf_prod:7.0f} {tf_13:7.0f} "
               f"{tf_13p:7.0f} {tf_13pr:7.0f} {tf_15p:7.0f} {tf_15pr:7.0f} "
               f"{tf_15pc:7.0f} {tf_16g:7.0f} {tf_16x:7.0f} {tf_17p:7.0f} "
               f"{tf_17pr:7.0f} {tf_19p:7.0f} {tf_19pr:7.0f} {tf_20p:7.0f} "
-              f"{tf_20pr:7.0f}")
+              f"{tf_20pr:7.0f} {tf_21p:7.0f} {tf_21pr:7.0f}")
 
 
 def main() -> None:
