@@ -112,7 +112,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int do_remap, int epi_staged) {
+    int c_is_bf16, int nbx, int nwg, int do_remap) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];  // one __shared__ object (G16 4a)
 #define AS(b) (lds + (b) * (BM * BK))
 #define BS(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -270,7 +270,7 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int do_remap, int epi_staged) {
+    int c_is_bf16, int nbx, int nwg, int do_remap) {
   __shared__ __bf16 lds[2 * (BM + BN) * BK];
 #define AS32(b) (lds + (b) * (BM * BK))
 #define BS32(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
@@ -435,12 +435,12 @@ __device__ __forceinline__ void stage_half(const __bf16* __restrict__ src,
   }
 }
 
-template <int ACT, bool HAS_BIAS, bool HAS_RES>
+template <int ACT, bool HAS_BIAS, bool HAS_RES, bool EPI_STAGED>
 __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     const __bf16* __restrict__ residual, long M, long N, long K,
-    int c_is_bf16, int nbx, int nwg, int order_mode, int epi_staged) {
+    int c_is_bf16, int nbx, int nwg, int order_mode) {
   // Round-2 structure: PERSISTENT multi-tile (grid = min(nwg, 256), one
   // WG per CU) around the round-1 8-phase glds schedule.  Each WG walks
   // tiles bid, bid+grid, ... and issues the NEXT tile's 12-glds
@@ -509,9 +509,14 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     /* level version of this, v12, regressed; the compile-time directive */ \
     /* does it without perturbing issue order elsewhere).  setprio       */ \
     /* dropped: it fenced the scheduler out of exactly this interleave.  */ \
-    _Pragma("unroll") for (int gi = 0; gi < 4; gi++) {                      \
-      __builtin_amdgcn_sched_group_barrier(0x8 /*MFMA*/, 4, 0);             \
+    /* pattern 3 of the sweep (profiles/r02_t19_sweep.log): lead with   */ \
+    /* 2 MFMAs, then weave [dsr, 2 MFMA, vmem] x6 — best at every ViT   */ \
+    /* shape (patch 1041, qkv 916, fc1 914, fc2 1056 TF vs 875-1009     */ \
+    /* for the coarser patterns).                                       */ \
+    __builtin_amdgcn_sched_group_barrier(0x8 /*MFMA*/, 2, 0);               \
+    _Pragma("unroll") for (int gi = 0; gi < 6; gi++) {                      \
       __builtin_amdgcn_sched_group_barrier(0x100 /*DS_READ*/, 1, 0);        \
+      __builtin_amdgcn_sched_group_barrier(0x8 /*MFMA*/, 2, 0);             \
       __builtin_amdgcn_sched_group_barrier(0x10 /*VMEM*/, 1, 0);            \
     }                                                                       \
     _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
@@ -643,7 +648,7 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
     // the scalar path, inside the embedding-cosine contract.
     const long crow_base = ebm + waveM * WM2 + 4 * (lane >> 4);
     const long ccol_base = ebn + waveN * WN2 + (lane & 15);
-    if (c_is_bf16 && epi_staged) {
+    if (c_is_bf16 && EPI_STAGED) {
       __bf16* slice = A2T(1) + wid * 2048;  // 4 KB per wave
       const long wrow0 = ebm + waveM * WM2;
       const long wcol0 = ebn + waveN * WN2;
@@ -817,18 +822,37 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   // residual (A/B: tower 582 -> 812 TF, bench 3184 -> 4188 clips/s,
   // profiles/r02_epi_ab.log), scalar for bare C=A*B^T launches where the
   // 4-pass staging is pure overhead (bare fc1 928 scalar vs 825 staged,
-  // profiles/r02_gemm_v20.log).  CC_GEMM_EPI=0|1 forces it for A/B.
+  // profiles/r02_gemm_v20.log).  A TEMPLATE parameter, not a runtime
+  // branch: carrying both epilogues in one instantiation spilled VGPRs
+  // in the rep loop and cost ~10% bare (profiles/r02_t19_sweep.log,
+  // prod column vs the single-epilogue tool kernels).
+  // CC_GEMM_EPI=0|1 forces it for A/B.
   static const int epi_env = [] {
     const char* e = getenv("CC_GEMM_EPI");
     return e ? atoi(e) : -1;
   }();
   const int epi_auto = (bias != nullptr || residual != nullptr || act != 0) ? 1 : 0;
-  const int epi_staged = epi_env >= 0 ? epi_env : epi_auto;
+  const bool epi_staged = (epi_env >= 0 ? epi_env : epi_auto) != 0;
 #define CC_LAUNCH_GEMM(KER, A_, HB, HR)                                       \
   hipLaunchKernelGGL((KER<A_, HB, HR>), grid, block, 0, (hipStream_t)stream,  \
                      (const __bf16*)A, (const __bf16*)B, C, bias,             \
                      (const __bf16*)residual, (long)M, (long)N, (long)K,      \
-                     c_dtype == 1 ? 1 : 0, nbx, nwg, remap, epi_staged)
+                     c_dtype == 1 ? 1 : 0, nbx, nwg, remap)
+#define CC_LAUNCH_T256(A_, HB, HR)                                            \
+  do {                                                                        \
+    if (epi_staged)                                                           \
+      hipLaunchKernelGGL((k_gemm_bf16_t256<A_, HB, HR, true>), grid, block,   \
+                         0, (hipStream_t)stream, (const __bf16*)A,            \
+                         (const __bf16*)B, C, bias, (const __bf16*)residual,  \
+                         (long)M, (long)N, (long)K, c_dtype == 1 ? 1 : 0,     \
+                         nbx, nwg, remap);                                    \
+    else                                                                      \
+      hipLaunchKernelGGL((k_gemm_bf16_t256<A_, HB, HR, false>), grid, block,  \
+                         0, (hipStream_t)stream, (const __bf16*)A,            \
+                         (const __bf16*)B, C, bias, (const __bf16*)residual,  \
+                         (long)M, (long)N, (long)K, c_dtype == 1 ? 1 : 0,     \
+                         nbx, nwg, remap);                                    \
+  } while (0)
 #define CC_DISPATCH_ACT(KER, A_)                                              \
   do {                                                                        \
     if (hb && hr) CC_LAUNCH_GEMM(KER, A_, true, true);                        \
@@ -844,9 +868,19 @@ extern "C" int cc_gemm_bf16_ex(const void* A, const void* B, void* C,
   } while (0)
   if (wide)
     CC_DISPATCH(k_gemm_bf16_w32);
-  else if (t256)
-    CC_DISPATCH(k_gemm_bf16_t256);
-  else
+  else if (t256) {
+#define CC_DISPATCH_ACT_T256(A_)                                              \
+  do {                                                                        \
+    if (hb && hr) CC_LAUNCH_T256(A_, true, true);                             \
+    else if (hb) CC_LAUNCH_T256(A_, true, false);                             \
+    else if (hr) CC_LAUNCH_T256(A_, false, true);                             \
+    else CC_LAUNCH_T256(A_, false, false);                                    \
+  } while (0)
+    if (act == 1) CC_DISPATCH_ACT_T256(1);
+    else if (act == 2) CC_DISPATCH_ACT_T256(2);
+    else CC_DISPATCH_ACT_T256(0);
+#undef CC_DISPATCH_ACT_T256
+  } else
     CC_DISPATCH(k_gemm_bf16);
 #undef CC_DISPATCH
 #undef CC_DISPATCH_ACT

@@ -1291,6 +1291,285 @@ extern "C" int cc_gemm_v21(const void* A, const void* B, void* C, long M,
   return hipGetLastError() == hipSuccess ? 0 : -1;
 }
 
+// ---- V22: T19 pattern sweep (template<int PAT>) ----
+// v13 showed register staging costs more than the glds path's vmcnt(0)
+// drain (704 vs 791 at patch64), but persistence + remap wins at the
+// skinny-N ViT shapes (fc1 784 vs 705).  v15 keeps the production
+// KTILE2 glds schedule verbatim and adds the persistent outer loop,
+// issuing the NEXT tile's 12-glds prologue BEFORE the epilogue stores
+// so its HBM latency hides under the C writeback.
+namespace {
+
+template <int PERSIST, int PAT>
+__global__ __launch_bounds__(512, 1) void k_gemm_v22(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    void* __restrict__ C, long M, long N, long K, int c_is_bf16, int nbx,
+    int nwg, int do_remap) {
+  __shared__ __bf16 lds[2 * (BM + BN) * BK];  // 128 KiB
+#define A22(b) (lds + (b) * (BM * BK))
+#define B22(b) (lds + 2 * (BM * BK) + (b) * (BN * BK))
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int waveM = wid >> 2, waveN = wid & 3;
+  const long KT = K / BK;  // caller guarantees even, >= 4
+  const long srow = wid * 16;
+  const int arow_base = waveM * WM + (lane & 15);
+  const int brow_base = waveN * WN + (lane & 15);
+  bf16x8 bfragT[NFR][2];
+  bf16x8 afrag[2][2];
+
+#define STAGE_A22(bmv, t, h)                                                \
+  stage_half15(A, K, (bmv) + (h) * 128 + srow, M, (t) * BK,                 \
+               A22((t) & 1) + ((h) * 128 + srow) * BK, lane)
+#define STAGE_B22(bnv, t, h)                                                \
+  stage_half15(B, K, (bnv) + (h) * 128 + srow, N, (t) * BK,                 \
+               B22((t) & 1) + ((h) * 128 + srow) * BK, lane)
+
+  // tile index for rep r of this WG under the three order modes:
+  //   0: orig = bid + r*grid                  (stride)
+  //   1: xcd-remap(orig) of mode 0            (stride + L2 affinity)
+  //   2: orig = chunk_start(bid) + r          (contiguous row-major chunk:
+  //      the A row band stays hot in L2 across ~nbx consecutive reps)
+  int chunk_start = 0, chunk_len = 1;
+  if (PERSIST && do_remap == 2) {
+    int q = nwg / (int)gridDim.x, r = nwg % (int)gridDim.x;
+    int b = (int)blockIdx.x;
+    chunk_len = q + (b < r ? 1 : 0);
+    chunk_start = b * q + (b < r ? b : r);
+  }
+  const int tiles =
+      PERSIST ? (do_remap == 2 ? chunk_len
+                               : (nwg + (int)gridDim.x - 1) / (int)gridDim.x)
+              : 1;
+  long bm = 0, bn = 0;
+  {
+    int orig = do_remap == 2 ? chunk_start : (int)blockIdx.x;
+    if (do_remap == 1) {
+      int q = nwg >> 3, r = nwg & 7;
+      int xcd = orig & 7, lid = orig >> 3;
+      orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+    }
+    bm = (long)(orig / nbx) * BM;
+    bn = (long)(orig % nbx) * BN;
+  }
+  // first prologue (full-latency; later ones hide under epilogues)
+  STAGE_A22(bm, 0, 0);
+  STAGE_A22(bm, 0, 1);
+  STAGE_B22(bn, 0, 0);
+  STAGE_B22(bn, 0, 1);
+  STAGE_B22(bn, 1, 0);
+  STAGE_B22(bn, 1, 1);
+
+  for (int rep = 0; rep < tiles; rep++) {
+    if (PERSIST && do_remap != 2 &&
+        (int)blockIdx.x + rep * (int)gridDim.x >= nwg)
+      break;
+    f32x4 acc[MFR][NFR] = {};
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+#define PHASE_MFMA22(q)                                                     \
+  do {                                                                      \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                      \
+    /* T19 dictation for this barrier region: weave the NEXT phase's 4  */ \
+    /* ds_reads (and the glds issues) between the 16 MFMAs instead of   */ \
+    /* after them, so their latency hides under the cluster.            */ \
+    if constexpr (PAT == 0) {                                               \
+      _Pragma("unroll") for (int gi = 0; gi < 4; gi++) {                    \
+        __builtin_amdgcn_sched_group_barrier(0x8, 4, 0);                    \
+        __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);                  \
+        __builtin_amdgcn_sched_group_barrier(0x10, 1, 0);                   \
+      }                                                                     \
+    } else if constexpr (PAT == 1) {                                        \
+      _Pragma("unroll") for (int gi = 0; gi < 8; gi++) {                    \
+        __builtin_amdgcn_sched_group_barrier(0x8, 2, 0);                    \
+        __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);                  \
+        __builtin_amdgcn_sched_group_barrier(0x10, 1, 0);                   \
+      }                                                                     \
+    } else if constexpr (PAT == 2) {                                        \
+      _Pragma("unroll") for (int gi = 0; gi < 4; gi++) {                    \
+        __builtin_amdgcn_sched_group_barrier(0x8, 4, 0);                    \
+        __builtin_amdgcn_sched_group_barrier(0x100, 2, 0);                  \
+      }                                                                     \
+      __builtin_amdgcn_sched_group_barrier(0x10, 4, 0);                     \
+    } else if constexpr (PAT == 3) {                                        \
+      __builtin_amdgcn_sched_group_barrier(0x8, 2, 0);                      \
+      _Pragma("unroll") for (int gi = 0; gi < 6; gi++) {                    \
+        __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);                  \
+        __builtin_amdgcn_sched_group_barrier(0x8, 2, 0);                    \
+        __builtin_amdgcn_sched_group_barrier(0x10, 1, 0);                   \
+      }                                                                     \
+    }                                                                       \
+    _Pragma("unroll") for (int g = 0; g < 2; g++) {                         \
+      _Pragma("unroll") for (int m = 0; m < 2; m++) {                       \
+        _Pragma("unroll") for (int n = 0; n < NFR; n++) {                   \
+          acc[2 * (q) + m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(    \
+              afrag[m][g], bfragT[n][g], acc[2 * (q) + m][n], 0, 0, 0);     \
+        }                                                                   \
+      }                                                                     \
+    }                                                                       \
+  } while (0)
+
+#define READ_A22(At, q)                                                     \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int m = 0; m < 2; m++) afrag[m][g] =             \
+        frag_read(At, arow_base + (q) * 32 + m * FRAG, k16);                \
+  }
+
+#define READ_B22(Bt)                                                        \
+  _Pragma("unroll") for (int g = 0; g < 2; g++) {                           \
+    const int k16 = (g << 2) + (lane >> 4);                                 \
+    _Pragma("unroll") for (int n = 0; n < NFR; n++) bfragT[n][g] =          \
+        frag_read(Bt, brow_base + n * FRAG, k16);                           \
+  }
+
+#define KTILE22(t)                                                          \
+  do {                                                                      \
+    const __bf16* At = A22((t) & 1);                                        \
+    const __bf16* Bt = B22((t) & 1);                                        \
+    READ_B22(Bt);                                                           \
+    READ_A22(At, 0);                                                        \
+    if ((t) + 1 < KT) STAGE_A22(bm, (t) + 1, 0);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA22(0);                                                        \
+    READ_A22(At, 1);                                                        \
+    if ((t) + 1 < KT) STAGE_A22(bm, (t) + 1, 1);                            \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA22(1);                                                        \
+    READ_A22(At, 2);                                                        \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA22(2);                                                        \
+    READ_A22(At, 3);                                                        \
+    /* COUNTED publish (replaces the per-tile vmcnt(0) drain): the next  */ \
+    /* tile's LDS reads need A(t+1) [issued ph1-2, 4 loads/wave] and     */ \
+    /* B(t+1) [issued one tile earlier, strictly older] landed.  After   */ \
+    /* issuing B(t+2)'s 4 loads here, the per-wave VMEM queue newest-    */ \
+    /* first is [B(t+2):4, A(t+1):4, <older, incl B(t+1)>]; vmcnt(4)     */ \
+    /* waits until only B(t+2) remains in flight => everything the next  */ \
+    /* tile reads has retired to LDS, while B(t+2) still spans the       */ \
+    /* boundary.  r01's nondeterministic counted variant waited BEFORE   */ \
+    /* the B(t+2) issues with exactly 4 outstanding — a no-op wait       */ \
+    /* (profiles/r01_t256_det2.log), which this placement fixes; the     */ \
+    /* 8-run multi-shape determinism screen guards the claim.            */ \
+    if ((t) + 2 < KT) {                                                     \
+      STAGE_B22(bn, (t) + 2, 0);                                            \
+      STAGE_B22(bn, (t) + 2, 1);                                            \
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                      \
+    } else {                                                                \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
+    }                                                                       \
+    __builtin_amdgcn_s_barrier();                                           \
+    PHASE_MFMA22(3);                                                        \
+  } while (0)
+
+    for (long it = 0; it < KT / 2; ++it) {
+      KTILE22(2 * it);
+      KTILE22(2 * it + 1);
+    }
+
+    const long ebm = bm, ebn = bn;
+    // advance to next tile and issue its prologue BEFORE the epilogue:
+    // the LDS buffers are fully consumed (every wave drained its reads
+    // via lgkmcnt(0) before its last barrier), but another wave may
+    // still be inside the last phase -> barrier first.
+    if (PERSIST && rep + 1 < tiles &&
+        (do_remap == 2 ||
+         (int)blockIdx.x + (rep + 1) * (int)gridDim.x < nwg)) {
+      __builtin_amdgcn_s_barrier();
+      int orig = do_remap == 2
+                     ? chunk_start + rep + 1
+                     : (int)blockIdx.x + (rep + 1) * (int)gridDim.x;
+      if (do_remap == 1) {
+        int q = nwg >> 3, r = nwg & 7;
+        int xcd = orig & 7, lid = orig >> 3;
+        orig = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + lid;
+      }
+      bm = (long)(orig / nbx) * BM;
+      bn = (long)(orig % nbx) * BN;
+      STAGE_A22(bm, 0, 0);
+      STAGE_A22(bm, 0, 1);
+      STAGE_B22(bn, 0, 0);
+      STAGE_B22(bn, 0, 1);
+      STAGE_B22(bn, 1, 0);
+      STAGE_B22(bn, 1, 1);
+    }
+
+    const long crow_base = ebm + waveM * WM + 4 * (lane >> 4);
+    const long ccol_base = ebn + waveN * WN + (lane & 15);
+    const bool interior = (ebm + BM <= M) && (ebn + BN <= N);
+    if (interior) {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int m = 0; m < MFR; m++) {
+#pragma unroll
+        for (int n = 0; n < NFR; n++) {
+          const long col = ccol_base + n * FRAG;
+          if (col >= N) continue;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const long row = crow_base + m * FRAG + r;
+            if (row >= M) continue;
+            if (c_is_bf16)
+              ((unsigned short*)C)[row * N + col] = bf16_rne(acc[m][n][r]);
+            else
+              ((float*)C)[row * N + col] = acc[m][n][r];
+          }
+        }
+      }
+    }
+  }
+#undef A22
+#undef B22
+#undef STAGE_A22
+#undef STAGE_B22
+#undef PHASE_MFMA22
+#undef READ_A22
+#undef READ_B22
+#undef KTILE22
+}
+
+}  // namespace
+
+extern "C" int cc_gemm_v22(const void* A, const void* B, void* C, long M,
+                           long N, long K, int c_is_bf16, int pat,
+                           int remap, unsigned long long stream) {
+  if (K % (2 * BK) != 0 || K / BK < 4) return -2;
+  int nbx = (int)((N + BN - 1) / BN);
+  int nby = (int)((M + BM - 1) / BM);
+  int nwg = nbx * nby;
+  int grid = nwg < 256 ? nwg : 256;
+#define L22(P)                                                             \
+  hipLaunchKernelGGL((k_gemm_v22<1, P>), dim3(grid), dim3(512), 0,         \
+                     (hipStream_t)stream, (const __bf16*)A,                \
+                     (const __bf16*)B, C, M, N, K, c_is_bf16, nbx, nwg,    \
+                     remap)
+  switch (pat) {
+    case 0: L22(0); break;
+    case 1: L22(1); break;
+    case 2: L22(2); break;
+    case 3: L22(3); break;
+    default: return -3;
+  }
+#undef L22
+  return hipGetLastError() == hipSuccess ? 0 : -1;
+}
 // ---- V16: v15 + per-XCD atomic tile queues ----
 // v15's static chunking leaves a straggler tail when nwg/grid has a big
 // fractional part (out64: 789/256 = 3.08 -> 21 WGs run a 4th tile while
