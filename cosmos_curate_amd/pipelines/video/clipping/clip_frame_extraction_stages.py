@@ -181,6 +181,7 @@ class ClipFrameExtractionStage(CuratorStage):
             stream = torch.cuda.current_stream(dev).cuda_stream
             n_sel = len(idx)
             rgb: torch.Tensor | None = None
+            filled = np.zeros(n_sel, dtype=bool)
             sess = hotpath.DecodeSession(0, int(info.codec))
             try:
                 def drain() -> None:
@@ -209,6 +210,7 @@ class ClipFrameExtractionStage(CuratorStage):
                             f.y, f.uv, 1, fh, fw, f.pitch,
                             rgb[slot].data_ptr(), oh, ow, stream,
                         ))
+                        filled[slot] = True
                         launched = True
                     if launched:
                         torch.cuda.current_stream(dev).synchronize()
@@ -223,8 +225,12 @@ class ClipFrameExtractionStage(CuratorStage):
                 sess.close()
         finally:
             d.close()
-        if rgb is None:
-            raise RuntimeError("decode produced no frames")
+        if rgb is None or not filled.all():
+            # a dropped/duplicate-pts frame would otherwise leave an
+            # UNINITIALIZED slot in the output — fail loudly instead
+            missing = int(n_sel - int(filled.sum()))
+            msg = f"decoder did not produce {missing}/{n_sel} sampled frames"
+            raise RuntimeError(msg)
         total = int(counts.sum())
         if total == n_sel and np.all(counts == 1):
             return rgb
@@ -300,6 +306,7 @@ class ClipFrameExtractionStage(CuratorStage):
                 name, stats = self._timer.log_stats()
                 task.stage_perf[name] = stats
         return tasks
+
 
 def vcn_slot_map(pts_ticks: "np.ndarray", sampled_idx: "np.ndarray") -> dict[int, int]:
     """Map pts tick -> output slot for sampled PRESENTATION indices.
