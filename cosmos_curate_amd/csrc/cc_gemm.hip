@@ -401,14 +401,17 @@ __global__ __launch_bounds__(256, 2) void k_gemm_bf16_w32(
 }
 
 // ---- 256x256 8-phase template kernel (guide "The 256² 8-phase
-// template", schedule derived + race-screened in tools/gemm_v10.hip,
-// measured profiles/r01_v10_b64.log) ----
+// template", schedule derived + race-screened through tools/gemm_v10
+// ... v22; round-2 ladder in DESIGN.md §9a) ----
 // 512 threads = 8 waves (2M x 4N), per-wave 128x64 = 8x4 frags; LDS
-// 128 KiB (1 WG/CU); raw s_barrier phases, counted vmcnt(4) at phases
-// 4/8 only, B fragments held in registers per K-tile.  Wins the
-// K>=1536 or N>=1536 shapes (patch/fc2 ~900, qkv/fc1 ~730-770 TF at
-// the batch-64 bench shapes vs ~700-830 for the 128² body) by halving
-// the per-launch A/B HBM re-read traffic, which bounds those shapes.
+// 128 KiB; PERSISTENT fleet (grid = min(nwg, 256)); raw s_barrier
+// phases; COUNTED vmcnt(4) publishes at tile tops (B(t+2) spans each
+// boundary in flight); T19 scheduler weave of next-phase ds_reads into
+// the MFMA clusters; template-selected epilogue (LDS-staged coalesced
+// for fused bias/act/residual, scalar otherwise).  Serves every shape
+// whose 256-tile grid fills the fleet (the ViT GEMM mix; batch-64
+// per-shape TF in profiles/r02_gemm_v21.log + the epilogue-on numbers
+// in profiles/r02_epi_ab2.log).
 constexpr int BM2 = 256, BN2 = 256;
 constexpr int WM2 = 128, WN2 = 64;
 constexpr int MFR2 = WM2 / FRAG;  // 8
@@ -497,8 +500,10 @@ __global__ __launch_bounds__(512, 1) void k_gemm_bf16_t256(
   for (int rep = 0; rep < tiles; rep++) {
     if ((int)blockIdx.x + rep * (int)gridDim.x >= nwg) break;
     f32x4 acc[MFR2][NFR2] = {};
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // prologue drain
-    __builtin_amdgcn_s_barrier();                     // (counted waits unsound: det2)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // rep-top drain: the
+    __builtin_amdgcn_s_barrier();  // one full drain left (covers the 12
+    // prologue glds AND the previous epilogue's in-flight stores; a
+    // counted form here would need codegen-dependent store counts)
 
 #define PHASE_MFMA2(q)                                                      \
   do {                                                                      \
