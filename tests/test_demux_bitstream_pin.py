@@ -122,3 +122,35 @@ def test_bitstream_parser_on_c_abi_packets():
             ticks.append(pts)
     _segment_order_check(ticks, bs.access_unit_pocs_idr(pkts),
                          "c-abi:" + path.name)
+
+
+def test_pin_catches_ctts_corruption():
+    """Negative control: corrupt one ctts entry of the real B-frame
+    stream and the POC pin must DETECT the divergence — evidence the
+    cross-check has teeth, not just vacuous agreement."""
+    path = IMAGEIO_RES / "cockatoo.mp4"
+    if not path.is_file():
+        pytest.skip("cockatoo fixture not present")
+    data = bytearray(path.read_bytes())
+    i = bytes(data).find(b"ctts")
+    assert i > 0, "fixture has no ctts box"
+    # ctts box: size(4) type(4) ver/flags(4) entry_count(4) then
+    # (sample_count, sample_offset) pairs — swap the offsets of the
+    # first two entries if distinct, else scale one up
+    import struct
+
+    entry0 = i + 12
+    c0, o0 = struct.unpack_from(">iI", data, entry0)  # noqa: F841
+    c1, o1 = struct.unpack_from(">iI", data, entry0 + 8)  # noqa: F841
+    _, off0 = struct.unpack_from(">II", data, entry0)
+    _, off1 = struct.unpack_from(">II", data, entry0 + 8)
+    if off0 != off1:
+        struct.pack_into(">I", data, entry0 + 4, off1)
+        struct.pack_into(">I", data, entry0 + 8 + 4, off0)
+    else:
+        struct.pack_into(">I", data, entry0 + 4, off0 + 1024)
+    trk = mp4_demux.parse_mp4(bytes(data))[0]
+    pkts = mp4_demux.annexb_packets(bytes(data), trk)
+    pocs_idr = bs.access_unit_pocs_idr(pkts)
+    with pytest.raises(AssertionError):
+        _segment_order_check(trk.pts, pocs_idr, "corrupted-ctts")
