@@ -79,3 +79,25 @@ def test_worker_pool_gloo_world2():
     assert by_rank[0][1] == 9 and by_rank[1][1] == 6
     assert by_rank[0][2] == by_rank[1][2] == 15  # SUM across ranks
     assert by_rank[0][3] == pytest.approx(0.2)  # MAX over ranks
+
+
+@pytest.mark.timeout(180)
+def test_worker_pool_gloo_world4():
+    """world_size=4 sharding (the 8-GPU weak-scaling shape at half
+    width): shard sizes 2/1/1/1 over 5 videos, whole-job aggregation."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29517
+    procs = [ctx.Process(target=_worker, args=(r, 4, port, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=160) for _ in range(4)]
+    for p in procs:
+        p.join(timeout=30)
+    by_rank = {r[0]: r for r in results}
+    for r in range(4):
+        assert by_rank[r][1] != "ERROR", by_rank[r]
+    # 5 videos x 3 clips, round-robin shards 2/1/1/1 -> 6/3/3/3 clips
+    assert sorted(by_rank[r][1] for r in range(4)) == [3, 3, 3, 6]
+    assert all(by_rank[r][2] == 15 for r in range(4))  # SUM across ranks
+    assert by_rank[0][3] == pytest.approx(0.4)  # MAX over ranks
