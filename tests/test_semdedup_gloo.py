@@ -108,3 +108,24 @@ def test_combine_keep_masks_world2():
     for p in ps:
         p.join(timeout=30)
     assert results[0] == results[1] == [False] * 10  # every row pruned by someone
+
+
+@pytest.mark.timeout(180)
+def test_kmeans_allreduce_world4():
+    """world_size=4: the dedup k-means collective at the 8-GPU shape's
+    half width — all four ranks converge to identical centroids."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker, args=(r, 4, 29523, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(4):
+        rank, cent, labels, shard = q.get(timeout=160)
+        assert not (isinstance(cent, str) and cent == "ERROR"), labels
+        res[rank] = (cent, labels, shard)
+    for p in procs:
+        p.join(timeout=30)
+    base = res[0][0]
+    for rank in range(1, 4):
+        np.testing.assert_allclose(res[rank][0], base, atol=1e-6)
