@@ -47,11 +47,11 @@ def _config_to_argv(path: pathlib.Path) -> tuple[str, list[str]]:
 
 def main(argv: list[str] | None = None):
     argv = list(sys.argv[1:] if argv is None else argv)
-    if not argv:
+    if not argv or argv[0] in {"-h", "--help"}:
         names = "|".join(_ENTRY_POINTS)
         print(f"usage: run_pipeline <{names}> [flags] | run_pipeline config.(json|yaml)",
               file=sys.stderr)
-        raise SystemExit(2)
+        raise SystemExit(0 if argv else 2)
     first = pathlib.Path(argv[0])
     if first.suffix in _CONFIG_EXTENSIONS:
         pipeline, rest = _config_to_argv(first)
