@@ -140,8 +140,6 @@ def test_pin_catches_ctts_corruption():
     import struct
 
     entry0 = i + 12
-    c0, o0 = struct.unpack_from(">iI", data, entry0)  # noqa: F841
-    c1, o1 = struct.unpack_from(">iI", data, entry0 + 8)  # noqa: F841
     _, off0 = struct.unpack_from(">II", data, entry0)
     _, off1 = struct.unpack_from(">II", data, entry0 + 8)
     if off0 != off1:
